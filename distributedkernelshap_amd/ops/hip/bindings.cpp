@@ -1,0 +1,109 @@
+// Torch bindings for the CDNA4 KernelSHAP kernels (kshap_kernels.hip).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+
+#include <cstdint>
+
+extern "C" void launch_fill_random_masks(
+    uint8_t* masks, int B, int S, int M, int ne, int n_random,
+    const float* cdf, const int* sizes, int n_sizes, int num_paired,
+    uint32_t seed, const int32_t* inst_ids, hipStream_t stream);
+
+extern "C" int launch_fused_predict_linear(
+    const uint8_t* masks, const float* diff, const float* base, const float* wbg,
+    float* ey, int B, int S, int M, int Mpad, int Npad, int n_out, int act,
+    hipStream_t stream);
+
+extern "C" void launch_synth_chunk(
+    const uint8_t* masks, const float* x, const float* bg, const int* col_group,
+    float* out, int b, int S, int M, int N, int D, int s_lo, int s_hi,
+    hipStream_t stream);
+
+extern "C" int launch_wls_solve(
+    const uint8_t* masks, const float* kw, const float* ey_adj,
+    const float* total, float* phi, int B, int S, int M, int n_out,
+    hipStream_t stream);
+
+namespace {
+
+#define CHECK_DEV(t) TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be a contiguous device tensor")
+
+hipStream_t current_stream() {
+    return c10::hip::getCurrentHIPStream().stream();
+}
+
+void fill_random_masks(
+    torch::Tensor masks, int64_t ne, int64_t n_random,
+    torch::Tensor cdf, torch::Tensor sizes, int64_t num_paired, int64_t seed,
+    torch::Tensor inst_ids) {
+    CHECK_DEV(masks); CHECK_DEV(cdf); CHECK_DEV(sizes); CHECK_DEV(inst_ids);
+    TORCH_CHECK(masks.dtype() == torch::kUInt8, "masks must be uint8");
+    TORCH_CHECK(masks.dim() == 3, "masks must be (B,S,M)");
+    TORCH_CHECK(inst_ids.dtype() == torch::kInt32, "inst_ids must be int32");
+    int B = masks.size(0), S = masks.size(1), M = masks.size(2);
+    TORCH_CHECK(inst_ids.size(0) == B, "inst_ids length");
+    TORCH_CHECK(M <= 64, "fill_random_masks supports M <= 64");
+    launch_fill_random_masks(
+        masks.data_ptr<uint8_t>(), B, S, M, (int)ne, (int)n_random,
+        cdf.data_ptr<float>(), sizes.data_ptr<int>(), (int)cdf.size(0),
+        (int)num_paired, (uint32_t)seed, inst_ids.data_ptr<int32_t>(),
+        current_stream());
+}
+
+void fused_predict_linear(
+    torch::Tensor masks, torch::Tensor diff, torch::Tensor base,
+    torch::Tensor wbg, torch::Tensor ey, int64_t act) {
+    CHECK_DEV(masks); CHECK_DEV(diff); CHECK_DEV(base); CHECK_DEV(wbg); CHECK_DEV(ey);
+    int B = masks.size(0), S = masks.size(1), M = masks.size(2);
+    int n_out = diff.size(1), Mpad = diff.size(2), Npad = diff.size(3);
+    TORCH_CHECK(ey.size(0) == B && ey.size(1) == S && ey.size(2) == n_out, "ey shape");
+    TORCH_CHECK(base.size(0) == n_out && base.size(1) == Npad, "base shape");
+    TORCH_CHECK(wbg.size(0) == Npad, "wbg shape");
+    int rc = launch_fused_predict_linear(
+        masks.data_ptr<uint8_t>(), diff.data_ptr<float>(), base.data_ptr<float>(),
+        wbg.data_ptr<float>(), ey.data_ptr<float>(), B, S, M, Mpad, Npad,
+        n_out, (int)act, current_stream());
+    TORCH_CHECK(rc == 0, "fused_predict_linear: unsupported shape (Mpad<=64, Npad%16==0, Npad<=128, n_out in {1,2,4})");
+}
+
+void synth_chunk(
+    torch::Tensor masks, torch::Tensor x, torch::Tensor bg,
+    torch::Tensor col_group, torch::Tensor out, int64_t b, int64_t s_lo,
+    int64_t s_hi) {
+    CHECK_DEV(masks); CHECK_DEV(x); CHECK_DEV(bg); CHECK_DEV(col_group); CHECK_DEV(out);
+    int S = masks.size(1), M = masks.size(2);
+    int N = bg.size(0), D = bg.size(1);
+    TORCH_CHECK(out.size(0) == (s_hi - s_lo) * N && out.size(1) == D, "out shape");
+    launch_synth_chunk(
+        masks.data_ptr<uint8_t>(), x.data_ptr<float>(), bg.data_ptr<float>(),
+        col_group.data_ptr<int>(), out.data_ptr<float>(), (int)b, S, M, N, D,
+        (int)s_lo, (int)s_hi, current_stream());
+}
+
+void wls_solve(
+    torch::Tensor masks, torch::Tensor kw, torch::Tensor ey_adj,
+    torch::Tensor total, torch::Tensor phi) {
+    CHECK_DEV(masks); CHECK_DEV(kw); CHECK_DEV(ey_adj); CHECK_DEV(total); CHECK_DEV(phi);
+    int B = masks.size(0), S = masks.size(1), M = masks.size(2);
+    int n_out = ey_adj.size(2);
+    TORCH_CHECK(phi.size(0) == B && phi.size(1) == M && phi.size(2) == n_out, "phi shape");
+    int rc = launch_wls_solve(
+        masks.data_ptr<uint8_t>(), kw.data_ptr<float>(), ey_adj.data_ptr<float>(),
+        total.data_ptr<float>(), phi.data_ptr<float>(), B, S, M, n_out,
+        current_stream());
+    TORCH_CHECK(rc == 0, "wls_solve: unsupported shape (2<=M<=64, n_out<=8)");
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("fill_random_masks", &fill_random_masks,
+          "Philox coalition sampling (K2)");
+    m.def("fused_predict_linear", &fused_predict_linear,
+          "MFMA fused mask@diff GEMM + activation + background reduce (K3-K6)");
+    m.def("synth_chunk", &synth_chunk,
+          "masked-background perturbation synthesis tile (K3')");
+    m.def("wls_solve", &wls_solve,
+          "batched constrained WLS Shapley solve (K7)");
+}

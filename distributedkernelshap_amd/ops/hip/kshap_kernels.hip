@@ -1,0 +1,568 @@
+// MI355X (gfx950) CDNA4 kernels for distributed KernelSHAP.
+//
+// Native implementation of the numeric inner loop the reference delegates to
+// shap 0.35.0 (reference call site explainers/kernel_shap.py:250; kernel
+// inventory SURVEY.md §2.4):
+//   K2  fill_random_masks      — counter-based (Philox4x32-10) coalition
+//                                sampling, complement-paired, wave-parallel
+//   K3/K4/K5/K6 fused_predict_linear — MFMA-tiled (v_mfma_f32_16x16x4_f32)
+//                                mask @ diff GEMM with LDS-staged mask /
+//                                background-partial tiles, fused activation
+//                                (none/sigmoid/softmax) and fused weighted
+//                                background reduction -> ey.  The masked-
+//                                background perturbation synthesis is folded
+//                                algebraically: for a linear predictor,
+//                                logits(synth[s,n]) = base[n] + sum_g
+//                                mask[s,g]*(x_part[g]-bg_part[n,g]), so the
+//                                207,200-row synth matrix never touches HBM.
+//   K3' synth_chunk            — explicit masked-background blend for the
+//                                arbitrary-(torch)-predictor path.
+//   K7  wls_solve              — batched constrained weighted-least-squares:
+//                                Gram build over S samples, in-LDS Cholesky,
+//                                back-substitution of the eliminated feature.
+//
+// All kernels are wave64 / LDS-tiled for CDNA4; fp32 compute throughout.
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define WAVE 64
+
+// ------------------------------------------------------------------------- //
+// Philox4x32-10 counter-based RNG (deterministic per (seed, instance, draw))
+// ------------------------------------------------------------------------- //
+
+struct Philox {
+    uint32_t ctr[4];
+    uint32_t key[2];
+    uint32_t buf[4];
+    int idx;
+
+    __device__ void init(uint32_t k0, uint32_t k1, uint32_t c0, uint32_t c1) {
+        key[0] = k0; key[1] = k1;
+        ctr[0] = c0; ctr[1] = c1; ctr[2] = 0; ctr[3] = 0;
+        idx = 4;
+    }
+    __device__ static void round_(uint32_t c[4], const uint32_t k[2]) {
+        const uint32_t M0 = 0xD2511F53u, M1 = 0xCD9E8D57u;
+        uint64_t p0 = (uint64_t)M0 * c[0];
+        uint64_t p1 = (uint64_t)M1 * c[2];
+        uint32_t h0 = (uint32_t)(p0 >> 32), l0 = (uint32_t)p0;
+        uint32_t h1 = (uint32_t)(p1 >> 32), l1 = (uint32_t)p1;
+        uint32_t n0 = h1 ^ c[1] ^ k[0];
+        uint32_t n1 = l1;
+        uint32_t n2 = h0 ^ c[3] ^ k[1];
+        uint32_t n3 = l0;
+        c[0] = n0; c[1] = n1; c[2] = n2; c[3] = n3;
+    }
+    __device__ void gen() {
+        uint32_t c[4] = {ctr[0], ctr[1], ctr[2], ctr[3]};
+        uint32_t k[2] = {key[0], key[1]};
+        const uint32_t B0 = 0x9E3779B9u, B1 = 0xBB67AE85u;
+#pragma unroll
+        for (int r = 0; r < 10; ++r) {
+            round_(c, k);
+            k[0] += B0; k[1] += B1;
+        }
+        buf[0] = c[0]; buf[1] = c[1]; buf[2] = c[2]; buf[3] = c[3];
+        ctr[2]++;                     // advance the stream
+        if (ctr[2] == 0) ctr[3]++;
+        idx = 0;
+    }
+    __device__ uint32_t next_u32() {
+        if (idx >= 4) gen();
+        return buf[idx++];
+    }
+    // uniform integer in [0, n) via mul-shift
+    __device__ uint32_t next_below(uint32_t n) {
+        return (uint32_t)(((uint64_t)next_u32() * n) >> 32);
+    }
+};
+
+// ------------------------------------------------------------------------- //
+// K2: random coalition masks.  One wave per instance; draws are complement-
+// paired; the wave prefix-scans per-draw row consumption (1 or 2 rows) so all
+// 64 lanes emit rows in parallel while preserving the sequential pair layout.
+// ------------------------------------------------------------------------- //
+
+__global__ void fill_random_masks_kernel(
+    uint8_t* __restrict__ masks,      // (B, S, M)
+    int B, int S, int M,
+    int ne,                           // enumerated rows (already filled)
+    int n_random,                     // rows to fill: [ne, ne+n_random)
+    const float* __restrict__ cdf,    // (n_sizes,) cumulative probs
+    const int* __restrict__ sizes,    // (n_sizes,) subset sizes
+    int n_sizes,
+    int num_paired,                   // sizes <= num_paired get a complement
+    uint32_t seed,
+    const int32_t* __restrict__ inst_ids)  // (B,) global instance index (RNG key)
+{
+    const int b = blockIdx.x;
+    if (b >= B) return;
+    const int lane = threadIdx.x & (WAVE - 1);
+
+    // local copies of the (tiny) size table
+    float lcdf[32];
+    int lsizes[32];
+    for (int i = 0; i < n_sizes && i < 32; ++i) { lcdf[i] = cdf[i]; lsizes[i] = sizes[i]; }
+
+    uint8_t* mrow_base = masks + (size_t)b * S * M;
+
+    int remaining = n_random;
+    int written = 0;
+    uint32_t iter = 0;
+    while (remaining > 0) {
+        Philox rng;
+        rng.init(seed, (uint32_t)inst_ids[b], iter, (uint32_t)lane | 0x52000000u);
+        // draw subset size from the residual kernel distribution
+        float u = (rng.next_u32() >> 8) * (1.0f / 16777216.0f);
+        int si = 0;
+        while (si < n_sizes - 1 && u > lcdf[si]) ++si;
+        int ssize = lsizes[si];
+        bool paired = ssize <= num_paired;
+
+        uint64_t pairmask = __ballot(paired);
+        uint64_t below = pairmask & ((1ull << lane) - 1ull);
+        int rows_before = lane + __popcll(below);
+        int total_rows = WAVE + __popcll(pairmask);
+
+        if (rows_before < remaining) {
+            // sample ssize distinct bits by rejection (ssize <= ceil((M-1)/2)
+            // so acceptance >= 1/2 per try)
+            uint64_t bits = 0ull;
+            int got = 0;
+            int guard = 0;
+            while (got < ssize && guard < 4096) {
+                uint32_t r = rng.next_below((uint32_t)M);
+                uint64_t bit = 1ull << r;
+                if (!(bits & bit)) { bits |= bit; ++got; }
+                ++guard;
+            }
+            uint8_t* row = mrow_base + (size_t)(ne + written + rows_before) * M;
+            for (int g = 0; g < M; ++g) row[g] = (uint8_t)((bits >> g) & 1ull);
+            if (paired && rows_before + 1 < remaining) {
+                uint8_t* crow = row + M;
+                for (int g = 0; g < M; ++g) crow[g] = (uint8_t)(1u - ((bits >> g) & 1ull));
+            }
+        }
+        int consumed = total_rows < remaining ? total_rows : remaining;
+        written += consumed;
+        remaining -= consumed;
+        ++iter;
+    }
+}
+
+extern "C" void launch_fill_random_masks(
+    uint8_t* masks, int B, int S, int M, int ne, int n_random,
+    const float* cdf, const int* sizes, int n_sizes, int num_paired,
+    uint32_t seed, const int32_t* inst_ids, hipStream_t stream)
+{
+    if (n_random <= 0 || B <= 0) return;
+    fill_random_masks_kernel<<<dim3(B), dim3(WAVE), 0, stream>>>(
+        masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired, seed,
+        inst_ids);
+}
+
+// ------------------------------------------------------------------------- //
+// K3-K6 fused: ey[b,s,o] = sum_n wbg[n] * act( base[o,n]
+//                               + sum_k mask[b,s,k] * diff[b,o,k,n] )
+//
+// MFMA mask@diff GEMM (v_mfma_f32_16x16x4_f32, A = 16 s-rows x 4 k,
+// B = 4 k x 16 n-cols, C 16x16 fp32) with activation + weighted background
+// reduction fused in the epilogue.  LDS row strides are chosen ≡16 mod 32
+// words so the two k-rows read by a 32-lane group land on disjoint bank
+// halves (conflict-free ds_read_b32).
+//
+// Template params: NOUT in {1,2,4}; ACT 0=none 1=sigmoid 2=softmax.
+// ------------------------------------------------------------------------- //
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define MAX_MPAD 64        // fused path supports up to 64 varying groups
+#define S_TILE 64          // s-rows per workgroup (4 waves x 16)
+
+template <int NOUT, int ACT>
+__global__ __launch_bounds__(256)
+void fused_predict_linear_kernel(
+    const uint8_t* __restrict__ masks,  // (B, S, M)
+    const float* __restrict__ diff,     // (B, NOUT, Mpad, Npad)
+    const float* __restrict__ base,     // (NOUT, Npad)
+    const float* __restrict__ wbg,      // (Npad)  0 for padding cols
+    float* __restrict__ ey,             // (B, S, NOUT)
+    int B, int S, int M, int Mpad, int Npad)
+{
+    const int n_stiles = (S + S_TILE - 1) / S_TILE;
+    const int b = blockIdx.x / n_stiles;
+    const int stile = blockIdx.x % n_stiles;
+    const int s0 = stile * S_TILE;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wave = tid >> 6;           // 0..3
+
+    // LDS: mask tile [Mpad][MSTRIDE], diff [NOUT][Mpad][NSTRIDE], base, wbg
+    const int MSTRIDE = S_TILE + 16;     // 80 ≡ 16 mod 32
+    extern __shared__ float lds[];
+    float* mask_lds = lds;                                   // Mpad*MSTRIDE
+    const int NSTRIDE_PAD = ((16 - (Npad & 31)) & 31);
+    const int NSTRIDE = Npad + NSTRIDE_PAD;                  // ≡16 mod 32
+    float* diff_lds = mask_lds + Mpad * MSTRIDE;             // NOUT*Mpad*NSTRIDE
+    float* base_lds = diff_lds + NOUT * Mpad * NSTRIDE;      // NOUT*Npad
+    float* wbg_lds = base_lds + NOUT * Npad;                 // Npad
+
+    // ---- stage mask tile (transposed [k][s]) -------------------------------
+    for (int idx = tid; idx < Mpad * S_TILE; idx += 256) {
+        int k = idx / S_TILE, s = idx % S_TILE;
+        float v = 0.0f;
+        if (k < M && s0 + s < S)
+            v = (float)masks[((size_t)b * S + s0 + s) * M + k];
+        mask_lds[k * MSTRIDE + s] = v;
+    }
+    // ---- stage diff / base / wbg ------------------------------------------
+    const float* dsrc = diff + (size_t)b * NOUT * Mpad * Npad;
+    for (int idx = tid; idx < NOUT * Mpad * Npad; idx += 256) {
+        int o = idx / (Mpad * Npad);
+        int rem = idx % (Mpad * Npad);
+        int k = rem / Npad, n = rem % Npad;
+        diff_lds[(o * Mpad + k) * NSTRIDE + n] = dsrc[idx];
+    }
+    for (int idx = tid; idx < NOUT * Npad; idx += 256) base_lds[idx] = base[idx];
+    for (int idx = tid; idx < Npad; idx += 256) wbg_lds[idx] = wbg[idx];
+    __syncthreads();
+
+    // ---- MFMA loop ---------------------------------------------------------
+    const int NT = Npad / 16;            // col tiles (Npad % 16 == 0)
+    const int swave = wave * 16;         // this wave's 16 s-rows
+    const int arow = lane & 15;          // A row (s) within tile
+    const int akcol = lane >> 4;         // A col (k) 0..3
+
+    // NT <= 8; the ct loop is fully unrolled with a wave-uniform guard so
+    // every acc index is compile-time constant (runtime-indexed ext_vector
+    // arrays are demoted to scratch — guide §5.4 rule 20)
+    f32x4 acc[8][NOUT];
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+        for (int o = 0; o < NOUT; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
+
+    for (int ks = 0; ks < Mpad; ks += 4) {
+        float a = mask_lds[(ks + akcol) * MSTRIDE + swave + arow];
+#pragma unroll
+        for (int ct = 0; ct < 8; ++ct) {
+            if (ct < NT) {
+#pragma unroll
+                for (int o = 0; o < NOUT; ++o) {
+                    float bv = diff_lds[(o * Mpad + ks + akcol) * NSTRIDE + ct * 16 + arow];
+                    acc[ct][o] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[ct][o], 0, 0, 0);
+                }
+            }
+        }
+    }
+
+    // ---- epilogue: activation + weighted reduction over n ------------------
+    // C/D map: col = lane&15, row = (lane>>4)*4 + reg
+    float partial[NOUT][4];
+#pragma unroll
+    for (int o = 0; o < NOUT; ++o)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) partial[o][r] = 0.0f;
+
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct) {
+        if (ct >= NT) break;
+        int n = ct * 16 + (lane & 15);
+        float wn = wbg_lds[n];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            float z[NOUT];
+#pragma unroll
+            for (int o = 0; o < NOUT; ++o) z[o] = acc[ct][o][r] + base_lds[o * Npad + n];
+            if (ACT == 1) {
+#pragma unroll
+                for (int o = 0; o < NOUT; ++o) z[o] = 1.0f / (1.0f + __expf(-z[o]));
+            } else if (ACT == 2) {
+                float mx = z[0];
+#pragma unroll
+                for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
+                float sum = 0.0f;
+#pragma unroll
+                for (int o = 0; o < NOUT; ++o) { z[o] = __expf(z[o] - mx); sum += z[o]; }
+                float inv = 1.0f / sum;
+#pragma unroll
+                for (int o = 0; o < NOUT; ++o) z[o] *= inv;
+            }
+#pragma unroll
+            for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * z[o];
+        }
+    }
+    // reduce over the 16 lanes of each row group (xor bits 0-3 stay in-group)
+#pragma unroll
+    for (int o = 0; o < NOUT; ++o)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            float v = partial[o][r];
+            v += __shfl_xor(v, 1);
+            v += __shfl_xor(v, 2);
+            v += __shfl_xor(v, 4);
+            v += __shfl_xor(v, 8);
+            partial[o][r] = v;
+        }
+    if ((lane & 15) == 0) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            int s = s0 + swave + (lane >> 4) * 4 + r;
+            if (s < S) {
+#pragma unroll
+                for (int o = 0; o < NOUT; ++o)
+                    ey[((size_t)b * S + s) * NOUT + o] = partial[o][r];
+            }
+        }
+    }
+}
+
+template <int NOUT>
+static void launch_fused_act(
+    const uint8_t* masks, const float* diff, const float* base, const float* wbg,
+    float* ey, int B, int S, int M, int Mpad, int Npad, int act, hipStream_t stream)
+{
+    const int n_stiles = (S + S_TILE - 1) / S_TILE;
+    dim3 grid(B * n_stiles), block(256);
+    const int MSTRIDE = S_TILE + 16;
+    const int NSTRIDE = Npad + ((16 - (Npad & 31)) & 31);
+    size_t lds = (size_t)(Mpad * MSTRIDE + NOUT * Mpad * NSTRIDE + NOUT * Npad + Npad) * 4;
+    switch (act) {
+        case 0:
+            fused_predict_linear_kernel<NOUT, 0><<<grid, block, lds, stream>>>(
+                masks, diff, base, wbg, ey, B, S, M, Mpad, Npad);
+            break;
+        case 1:
+            fused_predict_linear_kernel<NOUT, 1><<<grid, block, lds, stream>>>(
+                masks, diff, base, wbg, ey, B, S, M, Mpad, Npad);
+            break;
+        default:
+            fused_predict_linear_kernel<NOUT, 2><<<grid, block, lds, stream>>>(
+                masks, diff, base, wbg, ey, B, S, M, Mpad, Npad);
+            break;
+    }
+}
+
+extern "C" int launch_fused_predict_linear(
+    const uint8_t* masks, const float* diff, const float* base, const float* wbg,
+    float* ey, int B, int S, int M, int Mpad, int Npad, int n_out, int act,
+    hipStream_t stream)
+{
+    if (Mpad > MAX_MPAD || Npad % 16 != 0 || Npad / 16 > 8) return -1;
+    switch (n_out) {
+        case 1: launch_fused_act<1>(masks, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 2: launch_fused_act<2>(masks, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 4: launch_fused_act<4>(masks, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
+        default: return -1;
+    }
+    return 0;
+}
+
+// ------------------------------------------------------------------------- //
+// K3': explicit masked-background synthesis for the torch-predictor path.
+// out[(s - s_lo)*N + n, d] = mask[b, s, group(d)] ? x[b, d] : bg[n, d]
+// Coalesced over d; one workgroup covers one (s, n) row-pair block.
+// ------------------------------------------------------------------------- //
+
+__global__ void synth_chunk_kernel(
+    const uint8_t* __restrict__ masks,   // (B, S, M)
+    const float* __restrict__ x,         // (B, D)
+    const float* __restrict__ bg,        // (N, D)
+    const int* __restrict__ col_group,   // (D)
+    float* __restrict__ out,             // ((s_hi-s_lo)*N, D)
+    int b, int S, int M, int N, int D, int s_lo, int s_hi)
+{
+    const int row = blockIdx.x;          // (s - s_lo) * N + n
+    const int nrows = (s_hi - s_lo) * N;
+    if (row >= nrows) return;
+    const int s = s_lo + row / N;
+    const int n = row % N;
+    const uint8_t* mrow = masks + ((size_t)b * S + s) * M;
+    const float* xrow = x + (size_t)b * D;
+    const float* brow = bg + (size_t)n * D;
+    float* orow = out + (size_t)row * D;
+    for (int d = threadIdx.x; d < D; d += blockDim.x) {
+        orow[d] = mrow[col_group[d]] ? xrow[d] : brow[d];
+    }
+}
+
+extern "C" void launch_synth_chunk(
+    const uint8_t* masks, const float* x, const float* bg, const int* col_group,
+    float* out, int b, int S, int M, int N, int D, int s_lo, int s_hi,
+    hipStream_t stream)
+{
+    int nrows = (s_hi - s_lo) * N;
+    int threads = D >= 256 ? 256 : (D >= 64 ? 64 : 32);
+    synth_chunk_kernel<<<dim3(nrows), dim3(threads), 0, stream>>>(
+        masks, x, bg, col_group, out, b, S, M, N, D, s_lo, s_hi);
+}
+
+// ------------------------------------------------------------------------- //
+// K7: batched constrained WLS solve.  One 256-thread workgroup per instance.
+//   etmp[s,i] = mask[s,i] - mask[s,last]           (i < M-1)
+//   ey2[s,o]  = eyAdj[s,o] - mask[s,last]*total[o]
+//   A = etmp^T diag(w) etmp;  r_o = etmp^T (w * ey2_o)
+//   A w = r (Cholesky);  phi[last] = total - sum(w)
+// Masks are packed to uint64 bitfields in LDS per 256-sample chunk; each
+// thread owns a strided set of Gram entries (upper triangle) / rhs entries.
+// ------------------------------------------------------------------------- //
+
+#define WLS_CHUNK 256
+#define WLS_MAX_M 64
+#define WLS_MAX_NOUT 8
+
+__global__ __launch_bounds__(256)
+void wls_solve_kernel(
+    const uint8_t* __restrict__ masks,   // (B, S, M)
+    const float* __restrict__ kw,        // (B, S)
+    const float* __restrict__ ey_adj,    // (B, S, n_out)
+    const float* __restrict__ total,     // (B, n_out)
+    float* __restrict__ phi,             // (B, M, n_out)
+    int B, int S, int M, int n_out)
+{
+    const int b = blockIdx.x;
+    if (b >= B) return;
+    const int tid = threadIdx.x;
+    const int mm = M - 1;
+    const int npairs = mm * (mm + 1) / 2;
+
+    __shared__ uint64_t pk[WLS_CHUNK];
+    __shared__ float wch[WLS_CHUNK];
+    __shared__ float eych[WLS_CHUNK][WLS_MAX_NOUT];
+    __shared__ float A[WLS_MAX_M * WLS_MAX_M];
+    __shared__ float rhs[WLS_MAX_M * WLS_MAX_NOUT];
+    __shared__ float tot_s[WLS_MAX_NOUT];
+
+    if (tid < n_out) tot_s[tid] = total[(size_t)b * n_out + tid];
+    __syncthreads();
+
+    // per-thread accumulators over its strided entries
+    float accA[8];        // up to 8 pairs per thread: npairs <= 2016, 256 thr
+    float accR[8];
+    const int pairs_per_thread = (npairs + 255) / 256;
+    const int rtot = mm * n_out;
+    const int r_per_thread = (rtot + 255) / 256;
+    for (int q = 0; q < 8; ++q) { accA[q] = 0.0f; accR[q] = 0.0f; }
+
+    const uint8_t* mbase = masks + (size_t)b * S * M;
+    const float* kwb = kw + (size_t)b * S;
+    const float* eyb = ey_adj + (size_t)b * S * n_out;
+
+    for (int c0 = 0; c0 < S; c0 += WLS_CHUNK) {
+        const int clen = min(WLS_CHUNK, S - c0);
+        __syncthreads();
+        if (tid < clen) {
+            const uint8_t* mrow = mbase + (size_t)(c0 + tid) * M;
+            uint64_t bits = 0ull;
+            for (int g = 0; g < M; ++g) bits |= ((uint64_t)(mrow[g] & 1)) << g;
+            pk[tid] = bits;
+            wch[tid] = kwb[c0 + tid];
+            float mlast = (float)((bits >> (M - 1)) & 1ull);
+            for (int o = 0; o < n_out; ++o)
+                eych[tid][o] = eyb[(size_t)(c0 + tid) * n_out + o] - mlast * tot_s[o];
+        }
+        __syncthreads();
+        // Gram entries
+        for (int q = 0; q < pairs_per_thread; ++q) {
+            int p = tid + q * 256;
+            if (p >= npairs) break;
+            // unrank upper-triangle pair (i <= j)
+            int i = 0, rem = p;
+            while (rem >= mm - i) { rem -= mm - i; ++i; }
+            int j = i + rem;
+            float s_acc = 0.0f;
+            for (int t = 0; t < clen; ++t) {
+                uint64_t bits = pk[t];
+                float ml = (float)((bits >> (M - 1)) & 1ull);
+                float ei = (float)((bits >> i) & 1ull) - ml;
+                float ej = (float)((bits >> j) & 1ull) - ml;
+                s_acc += wch[t] * ei * ej;
+            }
+            accA[q] += s_acc;
+        }
+        // rhs entries
+        for (int q = 0; q < r_per_thread; ++q) {
+            int p = tid + q * 256;
+            if (p >= rtot) break;
+            int i = p / n_out, o = p % n_out;
+            float s_acc = 0.0f;
+            for (int t = 0; t < clen; ++t) {
+                uint64_t bits = pk[t];
+                float ml = (float)((bits >> (M - 1)) & 1ull);
+                float ei = (float)((bits >> i) & 1ull) - ml;
+                s_acc += wch[t] * ei * eych[t][o];
+            }
+            accR[q] += s_acc;
+        }
+    }
+    __syncthreads();
+    // scatter accumulators into LDS A (symmetric) and rhs
+    for (int q = 0; q < pairs_per_thread; ++q) {
+        int p = tid + q * 256;
+        if (p >= npairs) break;
+        int i = 0, rem = p;
+        while (rem >= mm - i) { rem -= mm - i; ++i; }
+        int j = i + rem;
+        A[i * mm + j] = accA[q];
+        A[j * mm + i] = accA[q];
+    }
+    for (int q = 0; q < r_per_thread; ++q) {
+        int p = tid + q * 256;
+        if (p >= rtot) break;
+        rhs[p] = accR[q];
+    }
+    __syncthreads();
+
+    // Cholesky factorisation (thread 0; mm <= 63, cold relative to Gram build)
+    if (tid == 0) {
+        for (int k = 0; k < mm; ++k) {
+            float d = A[k * mm + k];
+            for (int t = 0; t < k; ++t) d -= A[k * mm + t] * A[k * mm + t];
+            d = sqrtf(fmaxf(d, 1e-20f));
+            A[k * mm + k] = d;
+            float inv = 1.0f / d;
+            for (int r = k + 1; r < mm; ++r) {
+                float v = A[r * mm + k];
+                for (int t = 0; t < k; ++t) v -= A[r * mm + t] * A[k * mm + t];
+                A[r * mm + k] = v * inv;
+            }
+        }
+    }
+    __syncthreads();
+    // triangular solves: one thread per output
+    if (tid < n_out) {
+        const int o = tid;
+        float y[WLS_MAX_M];
+        for (int i = 0; i < mm; ++i) {
+            float v = rhs[i * n_out + o];
+            for (int t = 0; t < i; ++t) v -= A[i * mm + t] * y[t];
+            y[i] = v / A[i * mm + i];
+        }
+        float w[WLS_MAX_M];
+        float sumw = 0.0f;
+        for (int i = mm - 1; i >= 0; --i) {
+            float v = y[i];
+            for (int t = i + 1; t < mm; ++t) v -= A[t * mm + i] * w[t];
+            w[i] = v / A[i * mm + i];
+        }
+        for (int i = 0; i < mm; ++i) sumw += w[i];
+        float* prow = phi + (size_t)b * M * n_out;
+        for (int i = 0; i < mm; ++i) prow[i * n_out + o] = w[i];
+        prow[(M - 1) * n_out + o] = tot_s[o] - sumw;
+    }
+}
+
+extern "C" int launch_wls_solve(
+    const uint8_t* masks, const float* kw, const float* ey_adj,
+    const float* total, float* phi, int B, int S, int M, int n_out,
+    hipStream_t stream)
+{
+    if (M < 2 || M > WLS_MAX_M || n_out > WLS_MAX_NOUT) return -1;
+    int mm = M - 1;
+    if (mm * (mm + 1) / 2 > 8 * 256) return -1;
+    wls_solve_kernel<<<dim3(B), dim3(256), 0, stream>>>(
+        masks, kw, ey_adj, total, phi, B, S, M, n_out);
+    return 0;
+}

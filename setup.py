@@ -15,7 +15,7 @@ try:
     from torch.utils import cpp_extension
 
     ext_modules = [
-        cpp_extension.CppExtension(
+        cpp_extension.CUDAExtension(
             name="distributedkernelshap_amd.ops._kshap_hip",
             sources=[
                 "distributedkernelshap_amd/ops/hip/bindings.cpp",

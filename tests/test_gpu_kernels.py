@@ -1,0 +1,305 @@
+"""GPU kernel numerics tests: each HIP kernel vs a plain PyTorch/numpy fp32
+reference of the same op, plus end-to-end GPU-vs-CPU engine equivalence."""
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+torch = pytest.importorskip("torch")
+if not torch.cuda.is_available():
+    pytest.skip("needs an MI355X", allow_module_level=True)
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from distributedkernelshap_amd.ops import load_extension
+
+    return load_extension()
+
+
+def test_extension_in_tree(ext):
+    # the loaded .so must be the in-tree build (driver checks loaded paths)
+    assert "distributedkernelshap_amd/ops" in ext.__file__
+
+
+# ----------------------------------------------------------------------- #
+# K2 sampler
+
+def test_fill_random_masks_properties(ext):
+    from distributedkernelshap_amd.core.sampler import plan_coalitions
+
+    plan = plan_coalitions(12)  # S=2072, ne=596
+    b, s, m = 4, plan.nsamples, 12
+    ne = plan.enum_masks.shape[0]
+    masks = torch.zeros(b, s, m, dtype=torch.uint8, device="cuda")
+    cdf = torch.tensor(
+        np.cumsum(plan.random_size_probs).astype(np.float32), device="cuda"
+    )
+    szs = torch.tensor(plan.random_sizes.astype(np.int32), device="cuda")
+    ids = torch.arange(b, dtype=torch.int32, device="cuda")
+    ext.fill_random_masks(masks, ne, plan.n_random, cdf, szs, 5, 0, ids)
+    mh = masks.cpu().numpy()
+    rnd = mh[:, ne:]
+    sizes = rnd.sum(axis=2)
+    # all rows filled with sizes from the residual range (draw 4-6, compl 6-8)
+    assert sizes.min() >= 4 and sizes.max() <= 8
+    # complement pairing: a paired draw (size<=5) is followed by its complement
+    for bi in range(b):
+        i = 0
+        while i < rnd.shape[1] - 1:
+            if sizes[bi, i] <= 5:
+                assert np.array_equal(rnd[bi, i + 1], 1 - rnd[bi, i])
+                i += 2
+            else:
+                i += 1
+    # determinism + per-instance keying
+    masks2 = torch.zeros_like(masks)
+    ext.fill_random_masks(masks2, ne, plan.n_random, cdf, szs, 5, 0, ids)
+    assert torch.equal(masks, masks2)
+    assert not np.array_equal(mh[0], mh[1])
+
+
+def test_fill_random_masks_sharding_invariance(ext):
+    """Row b of a batch keyed [0..B) equals row 0 of a batch keyed [b] —
+    the property the distributed shards rely on."""
+    from distributedkernelshap_amd.core.sampler import plan_coalitions
+
+    plan = plan_coalitions(12)
+    ne = plan.enum_masks.shape[0]
+    cdf = torch.tensor(
+        np.cumsum(plan.random_size_probs).astype(np.float32), device="cuda"
+    )
+    szs = torch.tensor(plan.random_sizes.astype(np.int32), device="cuda")
+    full = torch.zeros(4, plan.nsamples, 12, dtype=torch.uint8, device="cuda")
+    ext.fill_random_masks(
+        full, ne, plan.n_random, cdf, szs, 5, 0,
+        torch.arange(4, dtype=torch.int32, device="cuda"),
+    )
+    single = torch.zeros(1, plan.nsamples, 12, dtype=torch.uint8, device="cuda")
+    ext.fill_random_masks(
+        single, ne, plan.n_random, cdf, szs, 5, 0,
+        torch.tensor([2], dtype=torch.int32, device="cuda"),
+    )
+    assert torch.equal(full[2], single[0])
+
+
+# ----------------------------------------------------------------------- #
+# K3-K6 fused predict
+
+def _fused_reference(masks, diff, base, wbg, act):
+    """Plain torch fp32 reference of the fused kernel."""
+    mf = masks.float()                                  # (B,S,M)
+    b, s, m = mf.shape
+    n_out, mpad, npad = diff.shape[1:]
+    logits = torch.einsum("bsk,bokn->bson", mf, diff[:, :, :m]) + base[None, None]
+    if act == 1:
+        p = torch.sigmoid(logits)
+    elif act == 2:
+        p = torch.softmax(logits, dim=2)                # over o (dim 2 of bson)
+    else:
+        p = logits
+    return torch.einsum("bson,n->bso", p, wbg)
+
+
+@pytest.mark.parametrize("act", [0, 1, 2])
+@pytest.mark.parametrize("n_out", [1, 2])
+def test_fused_predict_linear_vs_torch(ext, act, n_out):
+    if n_out == 1 and act == 2:
+        pytest.skip("softmax needs n_out >= 2")
+    g = torch.Generator(device="cuda").manual_seed(42)
+    b, s, m, mpad, npad = 3, 200, 12, 12, 112
+    n = 100
+    masks = (torch.rand(b, s, m, generator=g, device="cuda") > 0.5).to(torch.uint8)
+    diff = torch.zeros(b, n_out, mpad, npad, device="cuda")
+    diff[:, :, :m, :n] = torch.randn(b, n_out, m, n, generator=g, device="cuda")
+    base = torch.zeros(n_out, npad, device="cuda")
+    base[:, :n] = torch.randn(n_out, n, generator=g, device="cuda")
+    wbg = torch.zeros(npad, device="cuda")
+    wbg[:n] = 1.0 / n
+    ey = torch.empty(b, s, n_out, device="cuda")
+    ext.fused_predict_linear(masks, diff, base, wbg, ey, act)
+    ref = _fused_reference(masks, diff, base, wbg, act)
+    assert torch.allclose(ey, ref, atol=2e-5, rtol=1e-4), (
+        (ey - ref).abs().max().item()
+    )
+
+
+def test_fused_predict_nonmultiple_shapes(ext):
+    """S not a multiple of 64, M not a multiple of 4, N not multiple of 16."""
+    g = torch.Generator(device="cuda").manual_seed(7)
+    b, s, m, n, n_out = 2, 130, 10, 37, 2
+    mpad, npad = 12, 48
+    masks = (torch.rand(b, s, m, generator=g, device="cuda") > 0.3).to(torch.uint8)
+    diff = torch.zeros(b, n_out, mpad, npad, device="cuda")
+    diff[:, :, :m, :n] = torch.randn(b, n_out, m, n, generator=g, device="cuda")
+    base = torch.zeros(n_out, npad, device="cuda")
+    base[:, :n] = torch.randn(n_out, n, generator=g, device="cuda")
+    wbg = torch.zeros(npad, device="cuda")
+    wbg[:n] = torch.rand(n, generator=g, device="cuda") + 0.1
+    ey = torch.empty(b, s, n_out, device="cuda")
+    ext.fused_predict_linear(masks, diff, base, wbg, ey, 2)
+    ref = _fused_reference(masks, diff, base, wbg, 2)
+    assert torch.allclose(ey, ref, atol=2e-5, rtol=1e-4)
+
+
+# ----------------------------------------------------------------------- #
+# K3' synth
+
+def test_synth_chunk_vs_numpy(ext):
+    g = torch.Generator(device="cuda").manual_seed(3)
+    b, s, mg, n, d = 2, 40, 5, 16, 23
+    masks = (torch.rand(b, s, mg, generator=g, device="cuda") > 0.5).to(torch.uint8)
+    x = torch.randn(b, d, generator=g, device="cuda")
+    bg = torch.randn(n, d, generator=g, device="cuda")
+    colg = torch.randint(0, mg, (d,), generator=g, device="cuda", dtype=torch.int32)
+    s_lo, s_hi = 10, 30
+    out = torch.empty((s_hi - s_lo) * n, d, device="cuda")
+    ext.synth_chunk(masks, x, bg, colg, out, 1, s_lo, s_hi)
+    mh, xh, bgh, cg = (t.cpu().numpy() for t in (masks, x, bg, colg))
+    expect = np.empty(((s_hi - s_lo) * n, d), dtype=np.float32)
+    for si in range(s_lo, s_hi):
+        for ni in range(n):
+            row = (si - s_lo) * n + ni
+            take_x = mh[1, si, cg].astype(bool)
+            expect[row] = np.where(take_x, xh[1], bgh[ni])
+    assert np.array_equal(out.cpu().numpy(), expect)
+
+
+# ----------------------------------------------------------------------- #
+# K7 WLS
+
+def test_wls_solve_vs_cpu(ext):
+    from distributedkernelshap_amd.core.solver import solve_wls
+
+    g = torch.Generator(device="cuda").manual_seed(11)
+    b, s, m, n_out = 5, 500, 12, 2
+    masks = (torch.rand(b, s, m, generator=g, device="cuda") > 0.5).to(torch.uint8)
+    # avoid degenerate all-equal columns
+    masks[:, :, 0] = (torch.rand(b, s, generator=g, device="cuda") > 0.3).to(torch.uint8)
+    kw = torch.rand(b, s, generator=g, device="cuda") + 0.05
+    ey = torch.randn(b, s, n_out, generator=g, device="cuda")
+    total = torch.randn(b, n_out, generator=g, device="cuda")
+    phi = torch.empty(b, m, n_out, device="cuda")
+    ext.wls_solve(masks, kw, ey, total, phi)
+    ph = phi.cpu().numpy()
+    for i in range(b):
+        ref = solve_wls(
+            masks[i].cpu().numpy(),
+            kw[i].double().cpu().numpy(),
+            ey[i].double().cpu().numpy(),
+            total[i].double().cpu().numpy(),
+        )
+        assert np.allclose(ph[i], ref, atol=5e-3, rtol=1e-3), np.abs(ph[i] - ref).max()
+        # constraint holds exactly by construction
+        assert np.allclose(ph[i].sum(axis=0), total[i].cpu().numpy(), atol=1e-4)
+
+
+def test_wls_solve_m63(ext):
+    from distributedkernelshap_amd.core.solver import solve_wls
+
+    g = torch.Generator(device="cuda").manual_seed(13)
+    b, s, m, n_out = 2, 3000, 63, 1
+    masks = (torch.rand(b, s, m, generator=g, device="cuda") > 0.5).to(torch.uint8)
+    kw = torch.rand(b, s, generator=g, device="cuda") + 0.05
+    ey = torch.randn(b, s, n_out, generator=g, device="cuda")
+    total = torch.randn(b, n_out, generator=g, device="cuda")
+    phi = torch.empty(b, m, n_out, device="cuda")
+    ext.wls_solve(masks, kw, ey, total, phi)
+    ph = phi.cpu().numpy()
+    for i in range(b):
+        ref = solve_wls(
+            masks[i].cpu().numpy(),
+            kw[i].double().cpu().numpy(),
+            ey[i].double().cpu().numpy(),
+            total[i].double().cpu().numpy(),
+        )
+        assert np.allclose(ph[i], ref, atol=2e-2, rtol=5e-3), np.abs(ph[i] - ref).max()
+
+
+# ----------------------------------------------------------------------- #
+# end-to-end engine
+
+def test_engine_gpu_vs_cpu_full_enumeration():
+    """M=10 -> nsamples capped at 2^10-2 (full enumeration, zero sampling
+    noise): GPU phi must match the fp64 CPU oracle to fp32 tolerance."""
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor
+
+    rng = np.random.Generator(np.random.Philox(key=[5, 5]))
+    d = 10
+    pred = LinearPredictor.random(d, 2, seed=5)
+    bg = rng.normal(size=(50, d))
+    X = rng.normal(size=(8, d))
+    cpu = KernelShapEngine(pred, bg, link="logit", seed=0, device="cpu")
+    gpu = KernelShapEngine(pred, bg, link="logit", seed=0, device="cuda")
+    sv_c = cpu.shap_values(X)
+    sv_g = gpu.shap_values(X)
+    for o in range(2):
+        assert np.allclose(sv_g[o], sv_c[o], atol=2e-4, rtol=1e-3), np.abs(
+            sv_g[o] - sv_c[o]
+        ).max()
+
+
+def test_engine_gpu_local_accuracy_adult():
+    """Adult-shaped config on GPU: sum(phi) == link(f(x)) - link(fnull)."""
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.core.links import logit
+    from distributedkernelshap_amd.models import LinearPredictor, make_adult_like
+
+    data = make_adult_like(n_instances=32, n_background=100, seed=0)
+    pred = LinearPredictor.random(data.X.shape[1], 2, seed=0)
+    eng = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cuda",
+    )
+    sv = eng.shap_values(data.X)
+    fx = logit(pred(data.X))
+    for o in range(2):
+        total = sv[o].sum(axis=1) + eng.expected_value[o]
+        assert np.abs(total - fx[:, o]).max() < 1e-3
+
+
+def test_engine_gpu_sharding_invariance():
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor, make_adult_like
+
+    data = make_adult_like(n_instances=8, n_background=50, seed=0)
+    pred = LinearPredictor.random(data.X.shape[1], 2, seed=0)
+    eng = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cuda",
+    )
+    full = eng.shap_values(data.X)
+    p1 = eng.shap_values(data.X[:4], instance_offset=0)
+    p2 = eng.shap_values(data.X[4:], instance_offset=4)
+    for o in range(2):
+        assert np.allclose(
+            full[o], np.concatenate([p1[o], p2[o]]), atol=1e-5
+        )
+
+
+def test_engine_gpu_torch_module_path():
+    """Arbitrary-predictor path (synth_chunk + torch MLP) vs CPU oracle on a
+    fully-enumerated config."""
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import TorchPredictor
+
+    torch.manual_seed(0)
+    d, n_out = 8, 2
+    module = torch.nn.Sequential(
+        torch.nn.Linear(d, 32), torch.nn.Tanh(), torch.nn.Linear(32, n_out),
+        torch.nn.Softmax(dim=-1),
+    ).double()
+    module = module.float()
+    pred = TorchPredictor(module, device="cuda")
+    rng = np.random.Generator(np.random.Philox(key=[9, 9]))
+    bg = rng.normal(size=(30, d))
+    X = rng.normal(size=(4, d))
+    cpu = KernelShapEngine(pred, bg, link="logit", seed=0, device="cpu")
+    gpu = KernelShapEngine(pred, bg, link="logit", seed=0, device="cuda")
+    sv_c = cpu.shap_values(X)
+    sv_g = gpu.shap_values(X)
+    for o in range(n_out):
+        assert np.allclose(sv_g[o], sv_c[o], atol=5e-4, rtol=5e-3), np.abs(
+            sv_g[o] - sv_c[o]
+        ).max()
