@@ -16,6 +16,8 @@ def load_extension():
     if _ext is not None:
         return _ext
     try:
+        import torch  # noqa: F401  (loads libc10/libtorch before the ext)
+
         from . import _kshap_hip  # type: ignore
 
         _ext = _kshap_hip
