@@ -165,6 +165,8 @@ class GpuKernelShap:
         b, s, m = masks.shape
         mpad = max(4, (m + 3) // 4 * 4)
         npad = (self.N + 15) // 16 * 16
+        masksT = t.empty(b, mpad, s, device=self.device)
+        self.ext.transpose_masks(masks, masksT)
         diff = t.zeros(b, self.n_out, mpad, npad, device=self.device)
         diff[:, :, :m, : self.N] = self._diff_tensor(X_dev, varying)
         base = t.zeros(self.n_out, npad, device=self.device)
@@ -173,7 +175,7 @@ class GpuKernelShap:
         wbg[: self.N] = self.bg_w
         ey = t.empty(b, s, self.n_out, device=self.device)
         self.ext.fused_predict_linear(
-            masks, diff.contiguous(), base, wbg, ey, self.linear["act"]
+            masksT, diff.contiguous(), base, wbg, ey, self.linear["act"], m
         )
         return ey
 
@@ -251,14 +253,12 @@ class GpuKernelShap:
 
         # bucket instances by varying-group pattern (benchmark case: 1 bucket)
         vmat = self._varying_matrix(X_dev)              # (B, G) bool, host
-        patterns: dict = {}
-        for i in range(b):
-            patterns.setdefault(vmat[i].tobytes(), []).append(i)
+        uniq, inverse = np.unique(vmat, axis=0, return_inverse=True)
 
-        for key, idxs in patterns.items():
-            varying = np.nonzero(np.frombuffer(key, dtype=bool))[0]
+        for u in range(uniq.shape[0]):
+            varying = np.nonzero(uniq[u])[0]
             m = len(varying)
-            ids = np.asarray(idxs)
+            ids = np.nonzero(inverse == u)[0]
             ids_t = t.tensor(ids, dtype=t.int64, device=self.device)
             if m == 0:
                 continue

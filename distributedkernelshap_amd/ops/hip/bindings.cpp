@@ -11,8 +11,12 @@ extern "C" void launch_fill_random_masks(
     uint32_t seed, const int32_t* inst_ids, hipStream_t stream);
 
 extern "C" int launch_fused_predict_linear(
-    const uint8_t* masks, const float* diff, const float* base, const float* wbg,
+    const float* masksT, const float* diff, const float* base, const float* wbg,
     float* ey, int B, int S, int M, int Mpad, int Npad, int n_out, int act,
+    hipStream_t stream);
+
+extern "C" void launch_transpose_masks(
+    const uint8_t* masks, float* masksT, int B, int S, int M, int Mpad,
     hipStream_t stream);
 
 extern "C" void launch_synth_chunk(
@@ -51,18 +55,31 @@ void fill_random_masks(
         current_stream());
 }
 
-void fused_predict_linear(
-    torch::Tensor masks, torch::Tensor diff, torch::Tensor base,
-    torch::Tensor wbg, torch::Tensor ey, int64_t act) {
-    CHECK_DEV(masks); CHECK_DEV(diff); CHECK_DEV(base); CHECK_DEV(wbg); CHECK_DEV(ey);
+void transpose_masks(torch::Tensor masks, torch::Tensor masksT) {
+    CHECK_DEV(masks); CHECK_DEV(masksT);
+    TORCH_CHECK(masks.dtype() == torch::kUInt8 && masksT.dtype() == torch::kFloat32);
     int B = masks.size(0), S = masks.size(1), M = masks.size(2);
+    int Mpad = masksT.size(1);
+    TORCH_CHECK(masksT.size(0) == B && masksT.size(2) == S && Mpad >= M, "masksT shape");
+    launch_transpose_masks(
+        masks.data_ptr<uint8_t>(), masksT.data_ptr<float>(), B, S, M, Mpad,
+        current_stream());
+}
+
+void fused_predict_linear(
+    torch::Tensor masksT, torch::Tensor diff, torch::Tensor base,
+    torch::Tensor wbg, torch::Tensor ey, int64_t act, int64_t m) {
+    CHECK_DEV(masksT); CHECK_DEV(diff); CHECK_DEV(base); CHECK_DEV(wbg); CHECK_DEV(ey);
+    TORCH_CHECK(masksT.dtype() == torch::kFloat32, "masksT must be f32 (B,Mpad,S)");
+    int B = masksT.size(0), S = masksT.size(2);
     int n_out = diff.size(1), Mpad = diff.size(2), Npad = diff.size(3);
+    TORCH_CHECK(masksT.size(1) == Mpad, "masksT/diff Mpad mismatch");
     TORCH_CHECK(ey.size(0) == B && ey.size(1) == S && ey.size(2) == n_out, "ey shape");
     TORCH_CHECK(base.size(0) == n_out && base.size(1) == Npad, "base shape");
     TORCH_CHECK(wbg.size(0) == Npad, "wbg shape");
     int rc = launch_fused_predict_linear(
-        masks.data_ptr<uint8_t>(), diff.data_ptr<float>(), base.data_ptr<float>(),
-        wbg.data_ptr<float>(), ey.data_ptr<float>(), B, S, M, Mpad, Npad,
+        masksT.data_ptr<float>(), diff.data_ptr<float>(), base.data_ptr<float>(),
+        wbg.data_ptr<float>(), ey.data_ptr<float>(), B, S, (int)m, Mpad, Npad,
         n_out, (int)act, current_stream());
     TORCH_CHECK(rc == 0, "fused_predict_linear: unsupported shape (Mpad<=64, Npad%16==0, Npad<=128, n_out in {1,2,4})");
 }
@@ -102,6 +119,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "Philox coalition sampling (K2)");
     m.def("fused_predict_linear", &fused_predict_linear,
           "MFMA fused mask@diff GEMM + activation + background reduce (K3-K6)");
+    m.def("transpose_masks", &transpose_masks,
+          "masks (B,S,M) u8 -> (B,Mpad,S) f32 for coalesced A staging (K2b)");
     m.def("synth_chunk", &synth_chunk,
           "masked-background perturbation synthesis tile (K3')");
     m.def("wls_solve", &wls_solve,

@@ -184,7 +184,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 template <int NOUT, int ACT>
 __global__ __launch_bounds__(256)
 void fused_predict_linear_kernel(
-    const uint8_t* __restrict__ masks,  // (B, S, M)
+    const float* __restrict__ masksT,   // (B, Mpad, S) zero-padded rows k>=M
     const float* __restrict__ diff,     // (B, NOUT, Mpad, Npad)
     const float* __restrict__ base,     // (NOUT, Npad)
     const float* __restrict__ wbg,      // (Npad)  0 for padding cols
@@ -209,13 +209,11 @@ void fused_predict_linear_kernel(
     float* base_lds = diff_lds + NOUT * Mpad * NSTRIDE;      // NOUT*Npad
     float* wbg_lds = base_lds + NOUT * Npad;                 // Npad
 
-    // ---- stage mask tile (transposed [k][s]) -------------------------------
+    // ---- stage mask tile ([k][s], coalesced from the transposed layout) ----
+    const float* msrc = masksT + (size_t)b * Mpad * S;
     for (int idx = tid; idx < Mpad * S_TILE; idx += 256) {
         int k = idx / S_TILE, s = idx % S_TILE;
-        float v = 0.0f;
-        if (k < M && s0 + s < S)
-            v = (float)masks[((size_t)b * S + s0 + s) * M + k];
-        mask_lds[k * MSTRIDE + s] = v;
+        mask_lds[k * MSTRIDE + s] = (s0 + s < S) ? msrc[(size_t)k * S + s0 + s] : 0.0f;
     }
     // ---- stage diff / base / wbg ------------------------------------------
     const float* dsrc = diff + (size_t)b * NOUT * Mpad * Npad;
@@ -321,7 +319,7 @@ void fused_predict_linear_kernel(
 
 template <int NOUT>
 static void launch_fused_act(
-    const uint8_t* masks, const float* diff, const float* base, const float* wbg,
+    const float* masksT, const float* diff, const float* base, const float* wbg,
     float* ey, int B, int S, int M, int Mpad, int Npad, int act, hipStream_t stream)
 {
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
@@ -332,29 +330,29 @@ static void launch_fused_act(
     switch (act) {
         case 0:
             fused_predict_linear_kernel<NOUT, 0><<<grid, block, lds, stream>>>(
-                masks, diff, base, wbg, ey, B, S, M, Mpad, Npad);
+                masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad);
             break;
         case 1:
             fused_predict_linear_kernel<NOUT, 1><<<grid, block, lds, stream>>>(
-                masks, diff, base, wbg, ey, B, S, M, Mpad, Npad);
+                masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad);
             break;
         default:
             fused_predict_linear_kernel<NOUT, 2><<<grid, block, lds, stream>>>(
-                masks, diff, base, wbg, ey, B, S, M, Mpad, Npad);
+                masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad);
             break;
     }
 }
 
 extern "C" int launch_fused_predict_linear(
-    const uint8_t* masks, const float* diff, const float* base, const float* wbg,
+    const float* masksT, const float* diff, const float* base, const float* wbg,
     float* ey, int B, int S, int M, int Mpad, int Npad, int n_out, int act,
     hipStream_t stream)
 {
     if (Mpad > MAX_MPAD || Npad % 16 != 0 || Npad / 16 > 8) return -1;
     switch (n_out) {
-        case 1: launch_fused_act<1>(masks, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
-        case 2: launch_fused_act<2>(masks, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
-        case 4: launch_fused_act<4>(masks, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 1: launch_fused_act<1>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 2: launch_fused_act<2>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 4: launch_fused_act<4>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
         default: return -1;
     }
     return 0;
@@ -554,6 +552,159 @@ void wls_solve_kernel(
     }
 }
 
+
+// ------------------------------------------------------------------------- //
+// K2b: transpose masks (B,S,M) u8 -> (B,Mpad,S) f32 so the predict kernel
+// stages its A-operand with coalesced float loads.
+// ------------------------------------------------------------------------- //
+
+__global__ void transpose_masks_kernel(
+    const uint8_t* __restrict__ masks,  // (B, S, M)
+    float* __restrict__ masksT,         // (B, Mpad, S)
+    int B, int S, int M, int Mpad)
+{
+    const int b = blockIdx.y;
+    const int s = blockIdx.x * blockDim.x + threadIdx.x;
+    if (s >= S) return;
+    const uint8_t* row = masks + ((size_t)b * S + s) * M;
+    float* out = masksT + (size_t)b * Mpad * S + s;
+    for (int k = 0; k < M; ++k) out[(size_t)k * S] = (float)row[k];
+    for (int k = M; k < Mpad; ++k) out[(size_t)k * S] = 0.0f;
+}
+
+extern "C" void launch_transpose_masks(
+    const uint8_t* masks, float* masksT, int B, int S, int M, int Mpad,
+    hipStream_t stream)
+{
+    dim3 grid((S + 255) / 256, B), block(256);
+    transpose_masks_kernel<<<grid, block, 0, stream>>>(masks, masksT, B, S, M, Mpad);
+}
+
+// ------------------------------------------------------------------------- //
+// K7-MFMA: Gram + rhs build on matrix cores for mm + n_out <= 16.
+// One 16x16x4 f32 MFMA accumulates BOTH the (mm x mm) Gram matrix and the
+// (mm x n_out) rhs per 4 samples:  A[i][k] = e[k][i],
+// B[k][j] = w[k] * (j < mm ? e[k][j] : ey2[k][j-mm]).
+// 4 waves split the sample (K) axis; partial tiles summed through LDS, then
+// the same Cholesky/solve epilogue as the generic kernel.
+// ------------------------------------------------------------------------- //
+
+__global__ __launch_bounds__(256)
+void wls_solve_mfma_kernel(
+    const uint8_t* __restrict__ masks,   // (B, S, M)
+    const float* __restrict__ kw,        // (B, S)
+    const float* __restrict__ ey_adj,    // (B, S, n_out)
+    const float* __restrict__ total,     // (B, n_out)
+    float* __restrict__ phi,             // (B, M, n_out)
+    int B, int S, int M, int n_out)
+{
+    const int b = blockIdx.x;
+    if (b >= B) return;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wv = tid >> 6;             // wave 0..3
+    const int mm = M - 1;
+    const int cols = mm + n_out;         // <= 16
+
+    __shared__ uint64_t pk[WLS_CHUNK];
+    __shared__ float wch[WLS_CHUNK];
+    __shared__ float eych[WLS_CHUNK][WLS_MAX_NOUT];
+    __shared__ float tileA[4][16][17];   // per-wave 16x16 (+1 pad)
+    __shared__ float A[16 * 16];
+    __shared__ float rhs[16 * WLS_MAX_NOUT];
+    __shared__ float tot_s[WLS_MAX_NOUT];
+
+    if (tid < n_out) tot_s[tid] = total[(size_t)b * n_out + tid];
+
+    const uint8_t* mbase = masks + (size_t)b * S * M;
+    const float* kwb = kw + (size_t)b * S;
+    const float* eyb = ey_adj + (size_t)b * S * n_out;
+
+    const int arow = lane & 15;          // i (Gram row)
+    const int akk = lane >> 4;           // k within the 4-sample micro-step
+
+    f32x4 acc = (f32x4){0, 0, 0, 0};
+
+    for (int c0 = 0; c0 < S; c0 += WLS_CHUNK) {
+        const int clen = min(WLS_CHUNK, S - c0);
+        __syncthreads();
+        if (tid < clen) {
+            const uint8_t* mrow = mbase + (size_t)(c0 + tid) * M;
+            uint64_t bits = 0ull;
+            for (int g = 0; g < M; ++g) bits |= ((uint64_t)(mrow[g] & 1)) << g;
+            pk[tid] = bits;
+            wch[tid] = kwb[c0 + tid];
+            float mlast = (float)((bits >> (M - 1)) & 1ull);
+            for (int o = 0; o < n_out; ++o)
+                eych[tid][o] = eyb[(size_t)(c0 + tid) * n_out + o] - mlast * tot_s[o];
+        }
+        __syncthreads();
+        // wave wv covers samples [wv*64, wv*64+64) of the chunk, 4 per step
+        const int base = wv * 64;
+        for (int ks = 0; ks < 64; ks += 4) {
+            int t = base + ks + akk;     // chunk-local sample id
+            float a = 0.0f, bv = 0.0f;
+            if (t < clen) {
+                uint64_t bits = pk[t];
+                float ml = (float)((bits >> (M - 1)) & 1ull);
+                float w = wch[t];
+                if (arow < mm) a = (float)((bits >> arow) & 1ull) - ml;
+                if (arow < mm) bv = w * ((float)((bits >> arow) & 1ull) - ml);
+                else if (arow < cols) bv = w * eych[t][arow - mm];
+            }
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+        }
+    }
+    // C/D map: col = lane&15, row = (lane>>4)*4 + r
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+        tileA[wv][(lane >> 4) * 4 + r][lane & 15] = acc[r];
+    __syncthreads();
+    for (int idx = tid; idx < 16 * 16; idx += 256) {
+        int i = idx / 16, j = idx % 16;
+        float v = tileA[0][i][j] + tileA[1][i][j] + tileA[2][i][j] + tileA[3][i][j];
+        if (j < mm) A[i * mm + j] = v;           // Gram
+        else if (j < cols && i < mm) rhs[i * n_out + (j - mm)] = v;  // rhs
+    }
+    __syncthreads();
+
+    if (tid == 0) {
+        for (int k = 0; k < mm; ++k) {
+            float d = A[k * mm + k];
+            for (int t = 0; t < k; ++t) d -= A[k * mm + t] * A[k * mm + t];
+            d = sqrtf(fmaxf(d, 1e-20f));
+            A[k * mm + k] = d;
+            float inv = 1.0f / d;
+            for (int r = k + 1; r < mm; ++r) {
+                float v = A[r * mm + k];
+                for (int t = 0; t < k; ++t) v -= A[r * mm + t] * A[k * mm + t];
+                A[r * mm + k] = v * inv;
+            }
+        }
+    }
+    __syncthreads();
+    if (tid < n_out) {
+        const int o = tid;
+        float y[16];
+        for (int i = 0; i < mm; ++i) {
+            float v = rhs[i * n_out + o];
+            for (int t = 0; t < i; ++t) v -= A[i * mm + t] * y[t];
+            y[i] = v / A[i * mm + i];
+        }
+        float w[16];
+        float sumw = 0.0f;
+        for (int i = mm - 1; i >= 0; --i) {
+            float v = y[i];
+            for (int t = i + 1; t < mm; ++t) v -= A[t * mm + i] * w[t];
+            w[i] = v / A[i * mm + i];
+        }
+        for (int i = 0; i < mm; ++i) sumw += w[i];
+        float* prow = phi + (size_t)b * M * n_out;
+        for (int i = 0; i < mm; ++i) prow[i * n_out + o] = w[i];
+        prow[(M - 1) * n_out + o] = tot_s[o] - sumw;
+    }
+}
+
 extern "C" int launch_wls_solve(
     const uint8_t* masks, const float* kw, const float* ey_adj,
     const float* total, float* phi, int B, int S, int M, int n_out,
@@ -562,7 +713,12 @@ extern "C" int launch_wls_solve(
     if (M < 2 || M > WLS_MAX_M || n_out > WLS_MAX_NOUT) return -1;
     int mm = M - 1;
     if (mm * (mm + 1) / 2 > 8 * 256) return -1;
-    wls_solve_kernel<<<dim3(B), dim3(256), 0, stream>>>(
-        masks, kw, ey_adj, total, phi, B, S, M, n_out);
+    if (mm + n_out <= 16) {
+        wls_solve_mfma_kernel<<<dim3(B), dim3(256), 0, stream>>>(
+            masks, kw, ey_adj, total, phi, B, S, M, n_out);
+    } else {
+        wls_solve_kernel<<<dim3(B), dim3(256), 0, stream>>>(
+            masks, kw, ey_adj, total, phi, B, S, M, n_out);
+    }
     return 0;
 }

@@ -110,6 +110,11 @@ def test_fused_predict_linear_vs_torch(ext, act, n_out):
     b, s, m, mpad, npad = 3, 200, 12, 12, 112
     n = 100
     masks = (torch.rand(b, s, m, generator=g, device="cuda") > 0.5).to(torch.uint8)
+    masksT = torch.empty(b, mpad, s, device="cuda")
+    ext.transpose_masks(masks, masksT)
+    # transpose correctness (K2b)
+    assert torch.equal(masksT[:, :m].permute(0, 2, 1), masks.float())
+    assert torch.all(masksT[:, m:] == 0)
     diff = torch.zeros(b, n_out, mpad, npad, device="cuda")
     diff[:, :, :m, :n] = torch.randn(b, n_out, m, n, generator=g, device="cuda")
     base = torch.zeros(n_out, npad, device="cuda")
@@ -117,7 +122,7 @@ def test_fused_predict_linear_vs_torch(ext, act, n_out):
     wbg = torch.zeros(npad, device="cuda")
     wbg[:n] = 1.0 / n
     ey = torch.empty(b, s, n_out, device="cuda")
-    ext.fused_predict_linear(masks, diff, base, wbg, ey, act)
+    ext.fused_predict_linear(masksT, diff, base, wbg, ey, act, m)
     ref = _fused_reference(masks, diff, base, wbg, act)
     assert torch.allclose(ey, ref, atol=2e-5, rtol=1e-4), (
         (ey - ref).abs().max().item()
@@ -130,6 +135,8 @@ def test_fused_predict_nonmultiple_shapes(ext):
     b, s, m, n, n_out = 2, 130, 10, 37, 2
     mpad, npad = 12, 48
     masks = (torch.rand(b, s, m, generator=g, device="cuda") > 0.3).to(torch.uint8)
+    masksT = torch.empty(b, mpad, s, device="cuda")
+    ext.transpose_masks(masks, masksT)
     diff = torch.zeros(b, n_out, mpad, npad, device="cuda")
     diff[:, :, :m, :n] = torch.randn(b, n_out, m, n, generator=g, device="cuda")
     base = torch.zeros(n_out, npad, device="cuda")
@@ -137,7 +144,7 @@ def test_fused_predict_nonmultiple_shapes(ext):
     wbg = torch.zeros(npad, device="cuda")
     wbg[:n] = torch.rand(n, generator=g, device="cuda") + 0.1
     ey = torch.empty(b, s, n_out, device="cuda")
-    ext.fused_predict_linear(masks, diff, base, wbg, ey, 2)
+    ext.fused_predict_linear(masksT, diff, base, wbg, ey, 2, m)
     ref = _fused_reference(masks, diff, base, wbg, 2)
     assert torch.allclose(ey, ref, atol=2e-5, rtol=1e-4)
 
