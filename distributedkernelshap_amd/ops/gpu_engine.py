@@ -26,6 +26,29 @@ from ..core.sampler import CoalitionPlan
 
 logger = logging.getLogger(__name__)
 
+import os as _os
+import time as _time
+
+_TIMING = _os.environ.get("KSHAP_TIMING", "") == "1"
+
+
+class _StageTimer:
+    """Optional per-stage wall timing (KSHAP_TIMING=1): each mark syncs the
+    device, so only use for diagnosis."""
+
+    def __init__(self, torch_mod, enabled):
+        self.t = torch_mod
+        self.enabled = enabled
+        self.last = _time.perf_counter() if enabled else 0.0
+
+    def mark(self, name):
+        if not self.enabled:
+            return
+        self.t.cuda.synchronize()
+        now = _time.perf_counter()
+        print(f"[kshap-timing] {name}: {(now - self.last) * 1e3:.2f} ms", flush=True)
+        self.last = now
+
 _EPS = 1e-7  # fp32 logit clamp
 
 
@@ -240,9 +263,11 @@ class GpuKernelShap:
         instance_offset: int = 0,
     ) -> List[np.ndarray]:
         t = self.torch
+        timer = _StageTimer(t, _TIMING)
         X = np.ascontiguousarray(X, dtype=np.float64)
         b = X.shape[0]
         X_dev = t.tensor(X, dtype=t.float32, device=self.device)
+        timer.mark("h2d")
 
         fx = self._predict_rows(X_dev)                  # (B, n_out)
         lfx = self._link(fx)
@@ -253,7 +278,9 @@ class GpuKernelShap:
 
         # bucket instances by varying-group pattern (benchmark case: 1 bucket)
         vmat = self._varying_matrix(X_dev)              # (B, G) bool, host
+        timer.mark("varying")
         uniq, inverse = np.unique(vmat, axis=0, return_inverse=True)
+        timer.mark("bucket")
 
         for u in range(uniq.shape[0]):
             varying = np.nonzero(uniq[u])[0]
@@ -268,6 +295,7 @@ class GpuKernelShap:
             plan = self.engine._plan(m, nsamples)
             gids = ids + instance_offset
             masks, kw = self._device_masks(plan, gids)
+            timer.mark("masks")
             sub_X = X_dev[ids_t]
             if self.linear is not None:
                 mpad = max(4, (m + 3) // 4 * 4)
@@ -278,6 +306,7 @@ class GpuKernelShap:
                     ey = self._ey_linear_torch(masks, sub_X, varying)
             else:
                 ey = self._ey_torch_module(masks, sub_X, varying)
+            timer.mark("predict")
             ey_adj = (self._link(ey) - lfnull[None, None, :]).contiguous()
             total = total_all[ids_t].contiguous()
             if self._l1_active(plan, l1_reg):
@@ -288,10 +317,12 @@ class GpuKernelShap:
                     self.ext.wls_solve(masks, kw, ey_adj, total, phi)
                 else:
                     phi = self._solve_torch(masks, kw, ey_adj, total)
+            timer.mark("wls")
             vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
             phi_full[ids_t[:, None], vidx_t[None, :]] = phi
 
         out = phi_full.double().cpu().numpy()
+        timer.mark("d2h")
         return [np.ascontiguousarray(out[:, :, o]) for o in range(self.n_out)]
 
     # ------------------------------------------------------------------ #
