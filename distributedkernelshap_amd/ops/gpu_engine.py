@@ -279,7 +279,12 @@ class GpuKernelShap:
         # bucket instances by varying-group pattern (benchmark case: 1 bucket)
         vmat = self._varying_matrix(X_dev)              # (B, G) bool, host
         timer.mark("varying")
-        uniq, inverse = np.unique(vmat, axis=0, return_inverse=True)
+        if self.n_groups <= 64:
+            keys = vmat @ (1 << np.arange(self.n_groups, dtype=np.uint64))
+            _, first, inverse = np.unique(keys, return_index=True, return_inverse=True)
+            uniq = vmat[first]
+        else:
+            uniq, inverse = np.unique(vmat, axis=0, return_inverse=True)
         timer.mark("bucket")
 
         for u in range(uniq.shape[0]):

@@ -179,7 +179,8 @@ extern "C" void launch_fill_random_masks(
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define MAX_MPAD 64        // fused path supports up to 64 varying groups
-#define S_TILE 64          // s-rows per workgroup (4 waves x 16)
+#define S_TILE 256         // s-rows per workgroup (4 sub-tiles of 64)
+#define S_SUB 64           // rows per sub-tile (4 waves x 16)
 
 template <int NOUT, int ACT>
 __global__ __launch_bounds__(256)
@@ -200,7 +201,7 @@ void fused_predict_linear_kernel(
     const int wave = tid >> 6;           // 0..3
 
     // LDS: mask tile [Mpad][MSTRIDE], diff [NOUT][Mpad][NSTRIDE], base, wbg
-    const int MSTRIDE = S_TILE + 16;     // 80 ≡ 16 mod 32
+    const int MSTRIDE = S_SUB + 16;      // 80 ≡ 16 mod 32
     extern __shared__ float lds[];
     float* mask_lds = lds;                                   // Mpad*MSTRIDE
     const int NSTRIDE_PAD = ((16 - (Npad & 31)) & 31);
@@ -209,13 +210,8 @@ void fused_predict_linear_kernel(
     float* base_lds = diff_lds + NOUT * Mpad * NSTRIDE;      // NOUT*Npad
     float* wbg_lds = base_lds + NOUT * Npad;                 // Npad
 
-    // ---- stage mask tile ([k][s], coalesced from the transposed layout) ----
+    // ---- stage diff / base / wbg once; the 4 s-subtiles reuse them --------
     const float* msrc = masksT + (size_t)b * Mpad * S;
-    for (int idx = tid; idx < Mpad * S_TILE; idx += 256) {
-        int k = idx / S_TILE, s = idx % S_TILE;
-        mask_lds[k * MSTRIDE + s] = (s0 + s < S) ? msrc[(size_t)k * S + s0 + s] : 0.0f;
-    }
-    // ---- stage diff / base / wbg ------------------------------------------
     const float* dsrc = diff + (size_t)b * NOUT * Mpad * Npad;
     for (int idx = tid; idx < NOUT * Mpad * Npad; idx += 256) {
         int o = idx / (Mpad * Npad);
@@ -225,6 +221,16 @@ void fused_predict_linear_kernel(
     }
     for (int idx = tid; idx < NOUT * Npad; idx += 256) base_lds[idx] = base[idx];
     for (int idx = tid; idx < Npad; idx += 256) wbg_lds[idx] = wbg[idx];
+
+    for (int sub = 0; sub < S_TILE / S_SUB; ++sub) {
+    const int ssub0 = s0 + sub * S_SUB;
+    if (ssub0 >= S) break;
+    // ---- stage mask sub-tile ([k][s], coalesced from transposed layout) ---
+    __syncthreads();   // mask_lds rewritten each sub-tile; diff stable
+    for (int idx = tid; idx < Mpad * S_SUB; idx += 256) {
+        int k = idx / S_SUB, s = idx % S_SUB;
+        mask_lds[k * MSTRIDE + s] = (ssub0 + s < S) ? msrc[(size_t)k * S + ssub0 + s] : 0.0f;
+    }
     __syncthreads();
 
     // ---- MFMA loop ---------------------------------------------------------
@@ -307,7 +313,7 @@ void fused_predict_linear_kernel(
     if ((lane & 15) == 0) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-            int s = s0 + swave + (lane >> 4) * 4 + r;
+            int s = ssub0 + swave + (lane >> 4) * 4 + r;
             if (s < S) {
 #pragma unroll
                 for (int o = 0; o < NOUT; ++o)
@@ -315,6 +321,7 @@ void fused_predict_linear_kernel(
             }
         }
     }
+    }  // sub-tile loop
 }
 
 template <int NOUT>
@@ -324,7 +331,7 @@ static void launch_fused_act(
 {
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     dim3 grid(B * n_stiles), block(256);
-    const int MSTRIDE = S_TILE + 16;
+    const int MSTRIDE = S_SUB + 16;
     const int NSTRIDE = Npad + ((16 - (Npad & 31)) & 31);
     size_t lds = (size_t)(Mpad * MSTRIDE + NOUT * Mpad * NSTRIDE + NOUT * Npad + Npad) * 4;
     switch (act) {
