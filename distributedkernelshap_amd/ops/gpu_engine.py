@@ -101,6 +101,19 @@ class GpuKernelShap:
 
         self.fnull = t.tensor(engine.fnull, dtype=t.float32, device=self.device)
         self._enum_cache: dict = {}
+        # persistent workspaces: identical shapes every call, so reusing the
+        # same device blocks avoids caching-allocator churn (sporadic ~90 ms
+        # hipMalloc stalls measured at B=2560)
+        self._ws: dict = {}
+
+    def _buf(self, name, shape, dtype=None):
+        t = self.torch
+        dtype = dtype or t.float32
+        buf = self._ws.get(name)
+        if buf is None or buf.shape != tuple(shape) or buf.dtype != dtype:
+            buf = t.empty(*shape, dtype=dtype, device=self.device)
+            self._ws[name] = buf
+        return buf
 
     # ------------------------------------------------------------------ #
 
@@ -156,7 +169,7 @@ class GpuKernelShap:
             self._enum_cache[key] = (enum, ew, cdf, szs)
         enum, ew, cdf, szs = self._enum_cache[key]
 
-        masks = t.empty(b, s, m, dtype=t.uint8, device=self.device)
+        masks = self._buf("masks", (b, s, m), t.uint8)
         masks[:, :ne] = enum[None]
         kw = t.empty(s, dtype=t.float32, device=self.device)
         kw[:ne] = ew
@@ -168,7 +181,8 @@ class GpuKernelShap:
                 int(self.engine.seed), ids,
             )
             kw[ne:] = plan.weight_left / plan.n_random
-        kwb = kw[None, :].expand(b, s).contiguous()
+        kwb = self._buf("kwb", (b, s))
+        kwb.copy_(kw[None, :].expand(b, s))
         return masks, kwb
 
     # ------------------------------------------------------------------ #
@@ -188,17 +202,18 @@ class GpuKernelShap:
         b, s, m = masks.shape
         mpad = max(4, (m + 3) // 4 * 4)
         npad = (self.N + 15) // 16 * 16
-        masksT = t.empty(b, mpad, s, device=self.device)
+        masksT = self._buf("masksT", (b, mpad, s))
         self.ext.transpose_masks(masks, masksT)
-        diff = t.zeros(b, self.n_out, mpad, npad, device=self.device)
+        diff = self._buf("diff", (b, self.n_out, mpad, npad))
+        diff.zero_()
         diff[:, :, :m, : self.N] = self._diff_tensor(X_dev, varying)
         base = t.zeros(self.n_out, npad, device=self.device)
         base[:, : self.N] = self.baseN.T
         wbg = t.zeros(npad, device=self.device)
         wbg[: self.N] = self.bg_w
-        ey = t.empty(b, s, self.n_out, device=self.device)
+        ey = self._buf("ey", (b, s, self.n_out))
         self.ext.fused_predict_linear(
-            masksT, diff.contiguous(), base, wbg, ey, self.linear["act"], m
+            masksT, diff, base, wbg, ey, self.linear["act"], m
         )
         return ey
 
@@ -312,12 +327,18 @@ class GpuKernelShap:
             else:
                 ey = self._ey_torch_module(masks, sub_X, varying)
             timer.mark("predict")
-            ey_adj = (self._link(ey) - lfnull[None, None, :]).contiguous()
+            # in-place link transform: ey is a workspace, not needed afterwards
+            if self.link_name == "identity":
+                ey_adj = ey.sub_(lfnull[None, None, :])
+            else:
+                ey.clamp_(_EPS, 1.0 - _EPS)
+                ey.log_().sub_(t.log1p(-t.exp(ey)))  # log(p/(1-p)) in place
+                ey_adj = ey.sub_(lfnull[None, None, :])
             total = total_all[ids_t].contiguous()
             if self._l1_active(plan, l1_reg):
                 phi = self._solve_host_l1(masks, kw, ey_adj, total, l1_reg)
             else:
-                phi = t.empty(len(ids), m, self.n_out, device=self.device)
+                phi = self._buf("phi", (len(ids), m, self.n_out))
                 if 2 <= m <= 64 and self.n_out <= 8:
                     self.ext.wls_solve(masks, kw, ey_adj, total, phi)
                 else:
