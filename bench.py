@@ -70,8 +70,17 @@ def main() -> None:
         device=device,
     )
 
+    if use_cuda:
+        X_in = torch.from_numpy(data.X.astype(np.float32))
+        try:
+            X_in = X_in.pin_memory()
+        except RuntimeError:
+            pass
+    else:
+        X_in = data.X
+
     def step() -> np.ndarray:
-        sv = engine.shap_values(X=data.X, instance_offset=rank * args.instances)
+        sv = engine.shap_values(X=X_in, instance_offset=rank * args.instances)
         # gather per-instance shap rows (class 0) to every rank, reference
         # order_result parity
         if is_distributed():

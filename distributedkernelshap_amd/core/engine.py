@@ -152,12 +152,13 @@ class KernelShapEngine:
         ``instance_offset`` keys the per-instance counter RNG so distributed
         shards reproduce the single-process result exactly.
         """
-        X = _as_2d(X).astype(np.float64)
-        b = X.shape[0]
         if self._gpu is not None:
+            # torch tensors (incl. pinned/device) pass through untouched
             return self._gpu.shap_values(
                 X, nsamples=nsamples, l1_reg=l1_reg, instance_offset=instance_offset
             )
+        X = _as_2d(X).astype(np.float64)
+        b = X.shape[0]
         fx = np.asarray(self.predictor(X))
         if fx.ndim == 1:
             fx = fx.reshape(-1, 1)

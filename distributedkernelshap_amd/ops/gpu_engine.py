@@ -175,7 +175,12 @@ class GpuKernelShap:
         kw[:ne] = ew
         if plan.n_random > 0:
             num_paired = int(np.floor((m - 1) / 2.0))
-            ids = t.tensor(inst_ids.astype(np.int32), device=self.device)
+            base0 = int(inst_ids[0])
+            if np.array_equal(inst_ids, np.arange(base0, base0 + b)):
+                # contiguous ids (the common bucket): build on device, no H2D
+                ids = t.arange(base0, base0 + b, dtype=t.int32, device=self.device)
+            else:
+                ids = t.tensor(inst_ids.astype(np.int32), device=self.device)
             self.ext.fill_random_masks(
                 masks, ne, plan.n_random, cdf, szs, num_paired,
                 int(self.engine.seed), ids,
@@ -279,9 +284,17 @@ class GpuKernelShap:
     ) -> List[np.ndarray]:
         t = self.torch
         timer = _StageTimer(t, _TIMING)
-        X = np.ascontiguousarray(X, dtype=np.float64)
-        b = X.shape[0]
-        X_dev = t.tensor(X, dtype=t.float32, device=self.device)
+        if t.is_tensor(X):
+            b = X.shape[0]
+            if X.is_cuda:
+                X_dev = X.float()
+            else:
+                X_dev = self._buf("X", (b, X.shape[1]))
+                X_dev.copy_(X, non_blocking=True)   # pinned source = async DMA
+        else:
+            X = np.ascontiguousarray(X, dtype=np.float64)
+            b = X.shape[0]
+            X_dev = t.tensor(X, dtype=t.float32, device=self.device)
         timer.mark("h2d")
 
         fx = self._predict_rows(X_dev)                  # (B, n_out)
