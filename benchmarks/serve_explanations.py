@@ -1,0 +1,128 @@
+"""HTTP serving benchmark (reference ``benchmarks/serve_explanations.py`` C8
+parity): start a uvicorn server with the dynamic-batching app, fan the 2,560
+test instances out as concurrent single-instance requests, time the whole
+run, pickle ``{'t_elapsed': [...]}`` with the reference's filename scheme.
+"""
+import argparse
+import concurrent.futures
+import json
+import logging
+import multiprocessing as mp
+import os
+import pickle
+import sys
+import time
+import timeit
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
+import numpy as np  # noqa: E402
+
+logging.basicConfig(level=logging.INFO)
+logger = logging.getLogger(__name__)
+
+
+def _server_main(host, port, max_batch_size, assets_dir, device):
+    import uvicorn
+
+    from distributedkernelshap_amd.serve import BatchKernelShapModel, create_app
+    from distributedkernelshap_amd.utils import load_data, load_model
+
+    data = load_data(assets_dir)
+    clf = load_model(os.path.join(assets_dir, "predictor.pkl"))
+    model = BatchKernelShapModel(
+        clf,
+        data.background,
+        {"link": "logit", "seed": 0, "device": device},
+        {"groups": data.groups, "group_names": data.group_names},
+    )
+    app = create_app(model, max_batch_size=max_batch_size)
+    uvicorn.run(app, host=host, port=port, log_level="warning")
+
+
+def distribute_requests(X, url, max_workers):
+    """Fan out one GET-style request per instance (reference
+    ``distribute_request``/``explain``, serve_explanations.py:96-139)."""
+    import httpx
+
+    instances = np.split(X, X.shape[0])
+
+    def post(x):
+        with httpx.Client() as client:
+            r = client.post(url, json={"array": x.tolist()}, timeout=120.0)
+            r.raise_for_status()
+            return r.text
+
+    with concurrent.futures.ThreadPoolExecutor(max_workers=max_workers) as pool:
+        return list(pool.map(post, instances))
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--replicas", type=int, default=1,
+                        help="kept for filename parity; one GPU replica "
+                             "saturates the node")
+    parser.add_argument("--max-batch-size", type=int, default=64)
+    parser.add_argument("--instances", type=int, default=2560)
+    parser.add_argument("--nruns", type=int, default=3)
+    parser.add_argument("--concurrency", type=int, default=64)
+    parser.add_argument("--host", default="127.0.0.1")
+    parser.add_argument("--port", type=int, default=8800)
+    parser.add_argument("--device", default="auto")
+    parser.add_argument("--assets-dir", default="assets")
+    parser.add_argument("--results-dir", default="results")
+    args = parser.parse_args()
+
+    from distributedkernelshap_amd.utils import get_filename, load_data
+
+    data = load_data(args.assets_dir)
+    X = data.X_test[: args.instances]
+    assert X.shape[0] == args.instances
+
+    ctx = mp.get_context("spawn")
+    server = ctx.Process(
+        target=_server_main,
+        args=(args.host, args.port, args.max_batch_size, args.assets_dir,
+              args.device),
+        daemon=True,
+    )
+    server.start()
+    url = f"http://{args.host}:{args.port}/explain"
+    # wait for readiness
+    import httpx
+
+    for _ in range(600):
+        try:
+            if httpx.get(f"http://{args.host}:{args.port}/healthz",
+                         timeout=2.0).status_code == 200:
+                break
+        except Exception:
+            time.sleep(0.5)
+    else:
+        raise RuntimeError("server did not become ready")
+
+    path = get_filename(
+        args.replicas, 0, serve=True, max_batch_size=args.max_batch_size,
+        results_dir=args.results_dir,
+    )
+    result = {"t_elapsed": []}
+    try:
+        for run in range(args.nruns):
+            logger.info("run %d/%d", run + 1, args.nruns)
+            t_start = timeit.default_timer()
+            responses = distribute_requests(X, url, args.concurrency)
+            t_elapsed = timeit.default_timer() - t_start
+            logger.info("Time elapsed: %.4f s (%d responses)",
+                        t_elapsed, len(responses))
+            sv = np.asarray(json.loads(responses[0])["data"]["shap_values"][0])
+            assert sv.shape[1] == len(data.groups)
+            result["t_elapsed"].append(t_elapsed)
+            with open(path, "wb") as f:
+                pickle.dump(result, f)
+    finally:
+        server.terminate()
+        server.join(timeout=10)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,117 @@
+"""Typed configuration (the reference scattered knobs over
+argparse CLIs, a ``distributed_opts`` dict and Makefile env vars — here one
+dataclass tree covers all tiers, loadable from TOML/dict/CLI).
+"""
+from __future__ import annotations
+
+import dataclasses
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Union
+
+__all__ = [
+    "EngineConfig",
+    "DistributedConfig",
+    "ServeConfig",
+    "BenchConfig",
+    "KernelConfig",
+    "Config",
+]
+
+
+@dataclass
+class EngineConfig:
+    """Algorithm-core knobs (ride-through kwargs of the reference,
+    ``explainers/kernel_shap.py:835-851``)."""
+
+    link: str = "logit"
+    seed: int = 0
+    nsamples: Optional[int] = None  # None -> 2*M + 2048
+    l1_reg: Union[str, int, float, None] = "auto"
+    device: str = "auto"
+    chunk_rows: int = 1 << 20
+
+
+@dataclass
+class DistributedConfig:
+    """Reference DISTRIBUTED_OPTS (``explainers/kernel_shap.py:210-214``)
+    plus the collective-backend tier."""
+
+    n_workers: Optional[int] = None
+    batch_size: int = 1
+    actor_cpu_fraction: float = 1.0
+    backend: str = "auto"  # nccl (RCCL) on GPU, gloo on CPU
+    mp_context: Optional[str] = None
+
+    def to_opts(self) -> Dict[str, Any]:
+        d = {
+            "n_workers": self.n_workers,
+            "batch_size": self.batch_size,
+            "actor_cpu_fraction": self.actor_cpu_fraction,
+        }
+        if self.mp_context:
+            d["mp_context"] = self.mp_context
+        return d
+
+
+@dataclass
+class ServeConfig:
+    host: str = "127.0.0.1"
+    port: int = 8800
+    replicas: int = 1
+    max_batch_size: int = 64
+    max_wait_ms: float = 2.0
+
+
+@dataclass
+class KernelConfig:
+    """HIP kernel dispatch knobs (tile sizes are compile-time; these select
+    code paths)."""
+
+    wls_mode: str = "auto"  # auto | mfma | generic | torch
+    fused_predict: bool = True
+    synth_chunk_rows: int = 1 << 19
+
+
+@dataclass
+class BenchConfig:
+    instances: int = 2560
+    background: int = 100
+    nruns: int = 5
+    steps: int = 5
+    warmup: int = 2
+    batch_sizes: List[int] = field(default_factory=lambda: [1, 5, 10])
+    results_dir: str = "results"
+    assets_dir: str = "assets"
+
+
+@dataclass
+class Config:
+    engine: EngineConfig = field(default_factory=EngineConfig)
+    distributed: DistributedConfig = field(default_factory=DistributedConfig)
+    serve: ServeConfig = field(default_factory=ServeConfig)
+    kernels: KernelConfig = field(default_factory=KernelConfig)
+    bench: BenchConfig = field(default_factory=BenchConfig)
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "Config":
+        cfg = cls()
+        for section, values in d.items():
+            if not hasattr(cfg, section):
+                raise KeyError(f"unknown config section '{section}'")
+            sub = getattr(cfg, section)
+            names = {f.name for f in dataclasses.fields(sub)}
+            for k, v in values.items():
+                if k not in names:
+                    raise KeyError(f"unknown key '{section}.{k}'")
+                setattr(sub, k, v)
+        return cfg
+
+    @classmethod
+    def from_toml(cls, path: str) -> "Config":
+        import tomli
+
+        with open(path, "rb") as f:
+            return cls.from_dict(tomli.load(f))
+
+    def to_dict(self) -> Dict[str, Any]:
+        return dataclasses.asdict(self)

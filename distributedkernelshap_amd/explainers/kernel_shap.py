@@ -321,6 +321,43 @@ class KernelShap(Explainer, FitMixin):
 
     # ------------------------------------------------------------------ #
 
+    @staticmethod
+    def _coerce_data(data, want_names: bool = True):
+        """Normalise supported input containers to (ndarray, names).
+
+        Reference ``_get_data`` singledispatch (``explainers/kernel_shap.py:
+        544-671``) registered shap DenseData, np.ndarray, scipy sparse,
+        DataFrame and Series; here every container collapses to a dense
+        float64 ndarray plus column names when the container carries them.
+        """
+        names = None
+        try:
+            import pandas as pd
+
+            if isinstance(data, pd.Series):
+                names = [str(data.name)] if data.name is not None else None
+                data = data.to_numpy().reshape(1, -1)
+            elif isinstance(data, pd.DataFrame):
+                names = [str(c) for c in data.columns]
+                data = data.to_numpy()
+        except ImportError:  # pragma: no cover
+            pass
+        try:
+            import scipy.sparse as sp
+
+            if sp.issparse(data):
+                logger.warning(
+                    "Sparse background/input densified; KernelSHAP perturbation "
+                    "synthesis operates on dense rows."
+                )
+                data = data.toarray()
+        except ImportError:  # pragma: no cover
+            pass
+        arr = np.asarray(data, dtype=np.float64)
+        if arr.ndim == 1:
+            arr = arr.reshape(1, -1)
+        return (arr, names) if want_names else arr
+
     def fit(
         self,
         background_data: np.ndarray,
@@ -335,10 +372,11 @@ class KernelShap(Explainer, FitMixin):
         (reference ``explainers/kernel_shap.py:697-808``).
 
         ``summarise_background``: False | True (subsample) | 'kmeans'.
+        Accepts ndarray, DataFrame, Series or scipy sparse input.
         """
-        background_data = np.asarray(background_data, dtype=np.float64)
-        if background_data.ndim == 1:
-            background_data = background_data.reshape(1, -1)
+        background_data, inferred_names = self._coerce_data(background_data)
+        if group_names is None and groups is None and inferred_names is not None:
+            group_names = inferred_names
 
         bg_weights = None
         summarised = False
@@ -421,9 +459,7 @@ class KernelShap(Explainer, FitMixin):
             raise TypeError(
                 "KernelShap explainer has not been fitted; call fit() first"
             )
-        X = np.asarray(X, dtype=np.float64)
-        if X.ndim == 1:
-            X = X.reshape(1, -1)
+        X = self._coerce_data(X, want_names=False)
         kwargs.pop("silent", None)
         shap_values = self._explainer.get_explanation(X, **kwargs)
         self.expected_value = (

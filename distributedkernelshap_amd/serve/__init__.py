@@ -1,0 +1,1 @@
+from .app import KernelShapModel, BatchKernelShapModel, create_app  # noqa: F401

@@ -76,3 +76,78 @@ def methdispatch(func):
 
     wrapper.register = dispatcher.register
     return wrapper
+
+
+# --------------------------------------------------------------------- #
+# dataset / model persistence (reference utils.py:124-188 downloaded from
+# GCS; here datasets are synthetic-generated and cached locally — no network)
+
+ASSETS_DIR = os.environ.get("KSHAP_ASSETS_DIR", "assets")
+
+
+def load_data(assets_dir: str = None):
+    """Return the benchmark dataset Bunch (X_train, y_train, X_test,
+    background, groups, group_names), generating + caching it on first use
+    (reference ``load_data``, utils.py:160-188)."""
+    import pickle
+
+    assets_dir = assets_dir or ASSETS_DIR
+    path = os.path.join(assets_dir, "data.pkl")
+    if os.path.exists(path):
+        with open(path, "rb") as f:
+            return pickle.load(f)
+    from ..models.synthetic import make_adult_like
+
+    data = make_adult_like(n_instances=2560, n_background=100, seed=0)
+    rng = np.random.Generator(np.random.Philox(key=[0, 0xDA7A]))
+    n_train = 30000
+    d = data.X.shape[1]
+    X_train = np.concatenate(
+        [make_adult_like(n_instances=n_train - 2560, n_background=1, seed=7).X,
+         data.X],
+        axis=0,
+    )
+    w_true = rng.normal(0.0, 0.7, size=d)
+    logits = X_train @ w_true + rng.normal(0, 0.5, size=n_train)
+    y_train = (logits > 0).astype(np.int64)
+    bunch = Bunch(
+        X_train=X_train,
+        y_train=y_train,
+        X_test=data.X,
+        y_test=(data.X @ w_true > 0).astype(np.int64),
+        background=data.background,
+        groups=data.groups,
+        group_names=data.group_names,
+        category_map=data.category_map,
+    )
+    os.makedirs(assets_dir, exist_ok=True)
+    with open(path, "wb") as f:
+        pickle.dump(bunch, f)
+    return bunch
+
+
+def load_model(path: str = None):
+    """Unpickle a predictor; sklearn LogisticRegression is wrapped into the
+    native LinearPredictor so the fused GPU path applies
+    (reference ``load_model``, utils.py:137-157)."""
+    import pickle
+
+    path = path or os.path.join(ASSETS_DIR, "predictor.pkl")
+    with open(path, "rb") as f:
+        obj = pickle.load(f)
+    from ..models.predictors import LinearPredictor
+
+    if isinstance(obj, LinearPredictor) or callable(getattr(obj, "linear_params", None)):
+        return obj
+    coef = getattr(obj, "coef_", None)
+    if coef is not None:
+        intercept = np.atleast_1d(obj.intercept_)
+        if coef.shape[0] == 1:  # binary sigmoid == softmax([0, z])
+            w = np.vstack([np.zeros_like(coef[0]), coef[0]])
+            b = np.array([0.0, float(intercept[0])])
+        else:
+            w, b = coef, intercept
+        return LinearPredictor(w, b, activation="softmax")
+    if not callable(obj):
+        raise TypeError(f"loaded object {type(obj)} is not a predictor")
+    return obj
