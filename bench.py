@@ -1,26 +1,95 @@
-"""Flagship benchmark: explanations/sec on the Adult-shaped KernelSHAP config.
+"""Flagship benchmark: explanations/sec for distributed KernelSHAP on MI355X.
 
-BASELINE.json metric: "explanations/sec (2560 inst, 100-sample background) at
-1/2/4/8 MI355X" — 12 feature groups, nsamples = 2*12 + 2048 = 2072, logit
-link, logistic-regression predictor, synthetic Adult-shaped data with
-random-init weights (no network). Reference floors: 1.47 expl/s
-sequential, 20.5 expl/s 32-worker node, 44.9 expl/s 56-worker k8s cluster.
+Default (--config adult) is the BASELINE.json headline metric:
+"explanations/sec (2560 inst, 100-sample background) at 1/2/4/8 MI355X" —
+12 feature groups, nsamples = 2*12 + 2048 = 2072, logit link, logistic-
+regression predictor, synthetic Adult-shaped data with random-init weights.
+Reference floors: 1.47 expl/s sequential, 20.5 expl/s
+32-worker node, 44.9 expl/s 56-worker k8s cluster.
 
-One *step* = explaining 2,560 instances per GPU (weak scaling: per-GPU work
-fixed as N grows; whole-job value = N * 2560 / step_time). Launched by the
-driver as  torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N ...
-(one rank per GPU over RCCL).
+Extra configs (BASELINE.json configs 3-5):
+  --config stress   256 features x 1000-sample background, nsamples=2^14
+                    (library-GEMM + batched torch WLS paths, HBM streaming)
+  --config mlp      64 features, MLP predictor (synth kernel + torch module)
+  --config resnet   ResNet-18 on 224x224 superpixel masks (large
+                    perturbation-batch predict path)
+
+One *step* = explaining ``--instances`` instances per GPU (weak scaling:
+per-GPU work fixed as N grows; whole-job value = N * instances / step_time).
+Launched by the driver as  torchrun --nnodes=1 --nproc-per-node N bench.py
+--gpus N ...  (one rank per GPU over RCCL).
 """
 from __future__ import annotations
 
 import argparse
 import json
-import os
 import time
 
 import numpy as np
 
 BASELINE_EXPL_PER_S = 20.5  # reference best single-node (125.05 s / 2560)
+
+CONFIG_DEFAULTS = {
+    "adult": {"instances": 2560, "background": 100},
+    "stress": {"instances": 64, "background": 1000, "nsamples": 2 ** 14},
+    "mlp": {"instances": 256, "background": 100},
+    "resnet": {"instances": 2, "background": 1},
+}
+
+
+def build_problem(cfg: str, args, rank: int):
+    """Returns (X, background, groups, group_names, predictor, nsamples,
+    model_desc)."""
+    from distributedkernelshap_amd.models import (
+        LinearPredictor,
+        TorchPredictor,
+        make_adult_like,
+        make_tabular,
+    )
+
+    if cfg == "adult":
+        data = make_adult_like(
+            n_instances=args.instances, n_background=args.background,
+            seed=1000 + rank,
+        )
+        pred = LinearPredictor.random(data.X.shape[1], 2, seed=0)
+        return (data.X, data.background, data.groups, data.group_names, pred,
+                None, "adult-logreg (12 groups, D=50, n_out=2, logit link)")
+    if cfg == "stress":
+        data = make_tabular(
+            n_features=256, n_instances=args.instances,
+            n_background=args.background, seed=1000 + rank,
+        )
+        pred = LinearPredictor.random(256, 2, seed=0)
+        return (data.X, data.background, data.groups, data.group_names, pred,
+                args.nsamples or 2 ** 14,
+                "stress-linear (256 groups, bg=1000, nsamples=2^14)")
+    if cfg == "mlp":
+        from distributedkernelshap_amd.models import make_predictor
+
+        data = make_tabular(
+            n_features=64, n_instances=args.instances,
+            n_background=args.background, seed=1000 + rank,
+        )
+        pred = make_predictor("mlp", 64, 2, seed=0, hidden=256, layers=2)
+        return (data.X, data.background, data.groups, data.group_names, pred,
+                None, "mlp-256x2 (64 groups, torch predictor)")
+    if cfg == "resnet":
+        import torch
+
+        from distributedkernelshap_amd.models.resnet import (
+            make_superpixel_problem,
+            resnet18,
+        )
+
+        X, bg, groups, names = make_superpixel_problem(
+            n_instances=args.instances, hw=224, patch=32, seed=1000 + rank
+        )
+        module = resnet18(num_classes=10, seed=0)
+        pred = TorchPredictor(module)
+        return (X, bg, groups, names, pred, args.nsamples,
+                "resnet18-superpixel (49 patches, 224x224)")
+    raise ValueError(cfg)
 
 
 def main() -> None:
@@ -28,15 +97,25 @@ def main() -> None:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=5)
     p.add_argument("--warmup", type=int, default=2)
-    p.add_argument("--instances", type=int, default=2560, help="instances per GPU")
-    p.add_argument("--background", type=int, default=100)
+    p.add_argument("--config", default="adult",
+                   choices=["adult", "stress", "mlp", "resnet"])
+    p.add_argument("--instances", type=int, default=None,
+                   help="instances per GPU (default per config)")
+    p.add_argument("--background", type=int, default=None)
+    p.add_argument("--nsamples", type=int, default=None)
     p.add_argument("--device", default="auto", choices=["auto", "cuda", "cpu"])
     args = p.parse_args()
+    defaults = CONFIG_DEFAULTS[args.config]
+    if args.instances is None:
+        args.instances = defaults["instances"]
+    if args.background is None:
+        args.background = defaults["background"]
+    if args.nsamples is None:
+        args.nsamples = defaults.get("nsamples")
 
     import torch
 
     from distributedkernelshap_amd.core.engine import KernelShapEngine
-    from distributedkernelshap_amd.models import LinearPredictor, make_adult_like
     from distributedkernelshap_amd.parallel import (
         allgather_rows,
         broadcast_array,
@@ -50,39 +129,47 @@ def main() -> None:
     )
     device = "cuda" if use_cuda else "cpu"
 
-    # per-rank synthetic shard (weak scaling: fixed per-GPU work); model
-    # weights + background are created on rank 0 and broadcast (RCCL/xGMI)
-    data = make_adult_like(
-        n_instances=args.instances, n_background=args.background, seed=1000 + rank
+    X, background, groups, group_names, pred, nsamples, model_desc = build_problem(
+        args.config, args, rank
     )
-    pred0 = LinearPredictor.random(data.X.shape[1], 2, seed=0)
-    W = broadcast_array(pred0.weights)
-    bias = broadcast_array(pred0.bias)
-    background = broadcast_array(data.background)
-    pred = LinearPredictor(W, bias)
+    # weights + background ship from rank 0 (RCCL broadcast over xGMI) — the
+    # reference's actor-constructor broadcast
+    from distributedkernelshap_amd.models import LinearPredictor
+
+    if isinstance(pred, LinearPredictor):
+        W = broadcast_array(pred.weights)
+        b_ = broadcast_array(pred.bias)
+        pred = LinearPredictor(W, b_, pred.activation)
+    background = broadcast_array(background)
 
     engine = KernelShapEngine(
-        pred,
-        background,
-        groups=data.groups,
-        link="logit",
-        seed=0,
-        device=device,
+        pred, background, groups=groups, link="logit", seed=0, device=device
     )
 
     if use_cuda:
-        X_in = torch.from_numpy(data.X.astype(np.float32))
+        X_in = torch.from_numpy(X.astype(np.float32))
         try:
             X_in = X_in.pin_memory()
         except RuntimeError:
             pass
     else:
-        X_in = data.X
+        X_in = X
+
+    ekw = {}
+    if nsamples is not None:
+        ekw["nsamples"] = nsamples
+    if args.config != "adult":
+        # high-dim configs sample a tiny fraction of 2^M subsets, which would
+        # trip shap's l1_reg='auto' LARS pre-selection (host-side, per
+        # instance); the benchmark measures the un-regularised WLS path
+        ekw["l1_reg"] = False
 
     def step() -> np.ndarray:
-        sv = engine.shap_values(X=X_in, instance_offset=rank * args.instances)
-        # gather per-instance shap rows (class 0) to every rank, reference
-        # order_result parity
+        sv = engine.shap_values(
+            X=X_in, instance_offset=rank * args.instances, **ekw
+        )
+        # gather per-instance shap rows (class 0) to every rank — the
+        # reference's order_result gather
         if is_distributed():
             counts = [args.instances] * world
             return allgather_rows(sv[0], counts)
@@ -105,7 +192,6 @@ def main() -> None:
     sync()
     elapsed = time.perf_counter() - t0
 
-    # max over ranks
     if is_distributed():
         import torch.distributed as dist
 
@@ -118,8 +204,13 @@ def main() -> None:
     ms_per_step = elapsed / args.steps * 1000.0
     value = n_gpus * args.instances * args.steps / elapsed
     if rank == 0:
+        metric = (
+            "explanations/sec (2560 inst, 100-sample background)"
+            if args.config == "adult"
+            else f"explanations/sec ({args.config} config)"
+        )
         result = {
-            "metric": "explanations/sec (2560 inst, 100-sample background)",
+            "metric": metric,
             "value": value,
             "unit": "explanations/s",
             "n_gpus": n_gpus,
@@ -128,14 +219,15 @@ def main() -> None:
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": value / BASELINE_EXPL_PER_S,
+            "vs_baseline": (value / BASELINE_EXPL_PER_S
+                            if args.config == "adult" else None),
             "dtype": "fp32",
             "data": "synthetic",
             "config": {
-                "model": "adult-logreg (12 groups, D=50, n_out=2, logit link)",
+                "model": model_desc,
                 "global_batch": n_gpus * args.instances,
                 "background": args.background,
-                "nsamples": 2072,
+                "nsamples": args.nsamples or "default(2M+2048)",
                 "parallelism": f"dp{n_gpus}",
                 "device": device,
                 "out_shape": list(out.shape),

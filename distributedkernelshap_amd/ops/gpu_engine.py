@@ -244,33 +244,59 @@ class GpuKernelShap:
             ey[:, lo:hi] = t.einsum("bsno,n->bso", p, self.bg_w)
         return ey
 
-    def _ey_torch_module(self, masks, X_dev, varying, chunk_rows=1 << 19):
+    def _ey_torch_module(self, masks, X_dev, varying, chunk_rows=None):
         """K3' synth + torch predictor + weighted mean (arbitrary-predictor
-        path). The synth tile never leaves the device."""
+        path). Synth tiles never leave the device; multiple
+        instances are packed per predictor call so the python/launch overhead
+        amortises over the 1M-instance MLP config."""
         t = self.torch
         b, s, m = masks.shape
+        if chunk_rows is None:
+            chunk_rows = max(1 << 19, self.N)
+            # cap the synth buffer at ~2 GB
+            chunk_rows = min(chunk_rows, max(self.N, (1 << 29) // max(1, self.D)))
         ey = t.empty(b, s, self.n_out, device=self.device)
-        s_chunk = max(1, chunk_rows // self.N)
-        buf = t.empty(s_chunk * self.N, self.D, device=self.device)
         # masks cover varying groups only; map non-varying columns to a
         # sentinel always-zero mask column m
         vmap = np.full(self.n_groups, m, dtype=np.int64)
         for i, g in enumerate(varying):
             vmap[g] = i
-        mfull = t.zeros(b, s, m + 1, dtype=t.uint8, device=self.device)
+        mfull = self._buf("mfull", (b, s, m + 1), t.uint8)
+        mfull.zero_()
         mfull[:, :, :m] = masks
         colg = t.tensor(
             vmap[self.engine._col_group].astype(np.int32), device=self.device
         )
-        for bi in range(b):
-            for lo in range(0, s, s_chunk):
-                hi = min(lo + s_chunk, s)
-                rows = (hi - lo) * self.N
+        rows_per_inst = s * self.N
+        if rows_per_inst <= chunk_rows:
+            # pack g instances per call
+            g_inst = max(1, chunk_rows // rows_per_inst)
+            buf = self._buf("synth", (g_inst * rows_per_inst, self.D))
+            for lo in range(0, b, g_inst):
+                hi = min(lo + g_inst, b)
+                rows = (hi - lo) * rows_per_inst
                 out = buf[:rows]
-                self.ext.synth_chunk(mfull, X_dev, self.bg, colg, out, bi, lo, hi)
+                self.ext.synth_chunk(mfull, X_dev, self.bg, colg, out, lo, hi, 0, s)
                 y = self._predict_rows(out)
-                y = y.view(hi - lo, self.N, self.n_out)
-                ey[bi, lo:hi] = t.einsum("cno,n->co", y, self.bg_w)
+                ey[lo:hi] = t.einsum(
+                    "csno,n->cso",
+                    y.view(hi - lo, s, self.N, self.n_out),
+                    self.bg_w,
+                )
+        else:
+            s_chunk = max(1, chunk_rows // self.N)
+            buf = self._buf("synth", (s_chunk * self.N, self.D))
+            for bi in range(b):
+                for lo in range(0, s, s_chunk):
+                    hi = min(lo + s_chunk, s)
+                    rows = (hi - lo) * self.N
+                    out = buf[:rows]
+                    self.ext.synth_chunk(
+                        mfull, X_dev, self.bg, colg, out, bi, bi + 1, lo, hi
+                    )
+                    y = self._predict_rows(out)
+                    y = y.view(hi - lo, self.N, self.n_out)
+                    ey[bi, lo:hi] = t.einsum("cno,n->co", y, self.bg_w)
         return ey
 
     # ------------------------------------------------------------------ #

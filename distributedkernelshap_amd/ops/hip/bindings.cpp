@@ -21,8 +21,8 @@ extern "C" void launch_transpose_masks(
 
 extern "C" void launch_synth_chunk(
     const uint8_t* masks, const float* x, const float* bg, const int* col_group,
-    float* out, int b, int S, int M, int N, int D, int s_lo, int s_hi,
-    hipStream_t stream);
+    float* out, int S, int M, int N, int D, int b_lo, int b_hi, int s_lo,
+    int s_hi, hipStream_t stream);
 
 extern "C" int launch_wls_solve(
     const uint8_t* masks, const float* kw, const float* ey_adj,
@@ -86,16 +86,17 @@ void fused_predict_linear(
 
 void synth_chunk(
     torch::Tensor masks, torch::Tensor x, torch::Tensor bg,
-    torch::Tensor col_group, torch::Tensor out, int64_t b, int64_t s_lo,
-    int64_t s_hi) {
+    torch::Tensor col_group, torch::Tensor out, int64_t b_lo, int64_t b_hi,
+    int64_t s_lo, int64_t s_hi) {
     CHECK_DEV(masks); CHECK_DEV(x); CHECK_DEV(bg); CHECK_DEV(col_group); CHECK_DEV(out);
     int S = masks.size(1), M = masks.size(2);
     int N = bg.size(0), D = bg.size(1);
-    TORCH_CHECK(out.size(0) == (s_hi - s_lo) * N && out.size(1) == D, "out shape");
+    TORCH_CHECK(out.size(0) == (b_hi - b_lo) * (s_hi - s_lo) * N && out.size(1) == D,
+                "out shape");
     launch_synth_chunk(
         masks.data_ptr<uint8_t>(), x.data_ptr<float>(), bg.data_ptr<float>(),
-        col_group.data_ptr<int>(), out.data_ptr<float>(), (int)b, S, M, N, D,
-        (int)s_lo, (int)s_hi, current_stream());
+        col_group.data_ptr<int>(), out.data_ptr<float>(), S, M, N, D,
+        (int)b_lo, (int)b_hi, (int)s_lo, (int)s_hi, current_stream());
 }
 
 void wls_solve(

@@ -376,14 +376,16 @@ __global__ void synth_chunk_kernel(
     const float* __restrict__ x,         // (B, D)
     const float* __restrict__ bg,        // (N, D)
     const int* __restrict__ col_group,   // (D)
-    float* __restrict__ out,             // ((s_hi-s_lo)*N, D)
-    int b, int S, int M, int N, int D, int s_lo, int s_hi)
+    float* __restrict__ out,             // ((b_hi-b_lo)*(s_hi-s_lo)*N, D)
+    int S, int M, int N, int D, int b_lo, int b_hi, int s_lo, int s_hi)
 {
-    const int row = blockIdx.x;          // (s - s_lo) * N + n
-    const int nrows = (s_hi - s_lo) * N;
+    const int srange = s_hi - s_lo;
+    const size_t row = blockIdx.x;       // ((b-b_lo)*srange + (s-s_lo))*N + n
+    const size_t nrows = (size_t)(b_hi - b_lo) * srange * N;
     if (row >= nrows) return;
-    const int s = s_lo + row / N;
     const int n = row % N;
+    const int s = s_lo + (row / N) % srange;
+    const int b = b_lo + row / ((size_t)N * srange);
     const uint8_t* mrow = masks + ((size_t)b * S + s) * M;
     const float* xrow = x + (size_t)b * D;
     const float* brow = bg + (size_t)n * D;
@@ -395,13 +397,13 @@ __global__ void synth_chunk_kernel(
 
 extern "C" void launch_synth_chunk(
     const uint8_t* masks, const float* x, const float* bg, const int* col_group,
-    float* out, int b, int S, int M, int N, int D, int s_lo, int s_hi,
-    hipStream_t stream)
+    float* out, int S, int M, int N, int D, int b_lo, int b_hi, int s_lo,
+    int s_hi, hipStream_t stream)
 {
-    int nrows = (s_hi - s_lo) * N;
+    size_t nrows = (size_t)(b_hi - b_lo) * (s_hi - s_lo) * N;
     int threads = D >= 256 ? 256 : (D >= 64 ? 64 : 32);
-    synth_chunk_kernel<<<dim3(nrows), dim3(threads), 0, stream>>>(
-        masks, x, bg, col_group, out, b, S, M, N, D, s_lo, s_hi);
+    synth_chunk_kernel<<<dim3((unsigned)nrows), dim3(threads), 0, stream>>>(
+        masks, x, bg, col_group, out, S, M, N, D, b_lo, b_hi, s_lo, s_hi);
 }
 
 // ------------------------------------------------------------------------- //
