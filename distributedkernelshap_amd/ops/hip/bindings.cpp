@@ -47,7 +47,7 @@ void fill_random_masks(
     TORCH_CHECK(inst_ids.dtype() == torch::kInt32, "inst_ids must be int32");
     int B = masks.size(0), S = masks.size(1), M = masks.size(2);
     TORCH_CHECK(inst_ids.size(0) == B, "inst_ids length");
-    TORCH_CHECK(M <= 64, "fill_random_masks supports M <= 64");
+    TORCH_CHECK(M <= 256, "fill_random_masks supports M <= 256");
     launch_fill_random_masks(
         masks.data_ptr<uint8_t>(), B, S, M, (int)ne, (int)n_random,
         cdf.data_ptr<float>(), sizes.data_ptr<int>(), (int)cdf.size(0),

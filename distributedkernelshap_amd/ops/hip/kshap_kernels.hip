@@ -85,6 +85,7 @@ struct Philox {
 // 64 lanes emit rows in parallel while preserving the sequential pair layout.
 // ------------------------------------------------------------------------- //
 
+template <int WORDS>   // bitset words: M <= 64*WORDS (WORDS <= 4)
 __global__ void fill_random_masks_kernel(
     uint8_t* __restrict__ masks,      // (B, S, M)
     int B, int S, int M,
@@ -129,20 +130,24 @@ __global__ void fill_random_masks_kernel(
         if (rows_before < remaining) {
             // sample ssize distinct bits by rejection (ssize <= ceil((M-1)/2)
             // so acceptance >= 1/2 per try)
-            uint64_t bits = 0ull;
+            uint64_t bits[WORDS];
+#pragma unroll
+            for (int w = 0; w < WORDS; ++w) bits[w] = 0ull;
             int got = 0;
             int guard = 0;
-            while (got < ssize && guard < 4096) {
+            while (got < ssize && guard < 65536) {
                 uint32_t r = rng.next_below((uint32_t)M);
-                uint64_t bit = 1ull << r;
-                if (!(bits & bit)) { bits |= bit; ++got; }
+                uint64_t bit = 1ull << (r & 63);
+                if (!(bits[r >> 6] & bit)) { bits[r >> 6] |= bit; ++got; }
                 ++guard;
             }
             uint8_t* row = mrow_base + (size_t)(ne + written + rows_before) * M;
-            for (int g = 0; g < M; ++g) row[g] = (uint8_t)((bits >> g) & 1ull);
+            for (int g = 0; g < M; ++g)
+                row[g] = (uint8_t)((bits[g >> 6] >> (g & 63)) & 1ull);
             if (paired && rows_before + 1 < remaining) {
                 uint8_t* crow = row + M;
-                for (int g = 0; g < M; ++g) crow[g] = (uint8_t)(1u - ((bits >> g) & 1ull));
+                for (int g = 0; g < M; ++g)
+                    crow[g] = (uint8_t)(1u - ((bits[g >> 6] >> (g & 63)) & 1ull));
             }
         }
         int consumed = total_rows < remaining ? total_rows : remaining;
@@ -158,9 +163,29 @@ extern "C" void launch_fill_random_masks(
     uint32_t seed, const int32_t* inst_ids, hipStream_t stream)
 {
     if (n_random <= 0 || B <= 0) return;
-    fill_random_masks_kernel<<<dim3(B), dim3(WAVE), 0, stream>>>(
-        masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired, seed,
-        inst_ids);
+    const int words = (M + 63) / 64;
+    switch (words) {
+        case 1:
+            fill_random_masks_kernel<1><<<dim3(B), dim3(WAVE), 0, stream>>>(
+                masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
+                seed, inst_ids);
+            break;
+        case 2:
+            fill_random_masks_kernel<2><<<dim3(B), dim3(WAVE), 0, stream>>>(
+                masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
+                seed, inst_ids);
+            break;
+        case 3:
+            fill_random_masks_kernel<3><<<dim3(B), dim3(WAVE), 0, stream>>>(
+                masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
+                seed, inst_ids);
+            break;
+        default:
+            fill_random_masks_kernel<4><<<dim3(B), dim3(WAVE), 0, stream>>>(
+                masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
+                seed, inst_ids);
+            break;
+    }
 }
 
 // ------------------------------------------------------------------------- //
