@@ -102,11 +102,8 @@ __global__ void fill_random_masks_kernel(
     if (b >= B) return;
     const int lane = threadIdx.x & (WAVE - 1);
 
-    // local copies of the (tiny) size table
-    float lcdf[32];
-    int lsizes[32];
-    for (int i = 0; i < n_sizes && i < 32; ++i) { lcdf[i] = cdf[i]; lsizes[i] = sizes[i]; }
-
+    // size table read via L2/L3 (up to ceil((M-1)/2) <= 128 entries; a
+    // fixed-size local copy silently truncates for wide M)
     uint8_t* mrow_base = masks + (size_t)b * S * M;
 
     int remaining = n_random;
@@ -118,8 +115,8 @@ __global__ void fill_random_masks_kernel(
         // draw subset size from the residual kernel distribution
         float u = (rng.next_u32() >> 8) * (1.0f / 16777216.0f);
         int si = 0;
-        while (si < n_sizes - 1 && u > lcdf[si]) ++si;
-        int ssize = lsizes[si];
+        while (si < n_sizes - 1 && u > cdf[si]) ++si;
+        int ssize = sizes[si];
         bool paired = ssize <= num_paired;
 
         uint64_t pairmask = __ballot(paired);
