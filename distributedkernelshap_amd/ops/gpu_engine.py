@@ -420,18 +420,21 @@ class GpuKernelShap:
 
     def _solve_torch(self, masks, kw, ey_adj, total):
         """Batched torch WLS (M > 64 stress configs): normal equations via
-        bmm + torch.linalg.solve."""
+        bmm + torch.linalg.solve, in fp64 — the Shapley kernel weights span
+        several orders of magnitude, so the Gram matrix at M~200 is too
+        ill-conditioned for fp32 normal equations (MI355X fp64 is cheap
+        relative to this cold path)."""
         t = self.torch
-        z = masks.float()
+        z = masks.double()
         last = z[:, :, -1:]
         etmp = z[:, :, :-1] - last                       # (b, s, m-1)
-        ey2 = ey_adj - last * total[:, None, :]
-        wz = etmp * kw[:, :, None]
+        ey2 = ey_adj.double() - last * total.double()[:, None, :]
+        wz = etmp * kw.double()[:, :, None]
         a = t.bmm(wz.transpose(1, 2), etmp)              # (b, m-1, m-1)
         r = t.bmm(wz.transpose(1, 2), ey2)               # (b, m-1, o)
         w = t.linalg.solve(a, r)
-        phi_last = total[:, None, :] - w.sum(dim=1, keepdim=True)
-        return t.cat([w, phi_last], dim=1)
+        phi_last = total.double()[:, None, :] - w.sum(dim=1, keepdim=True)
+        return t.cat([w, phi_last], dim=1).float()
 
     def _solve_host_l1(self, masks, kw, ey_adj, total, l1_reg):
         """Cold path: l1 feature selection + solve on host, per instance."""
