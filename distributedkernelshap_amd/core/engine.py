@@ -121,6 +121,16 @@ class KernelShapEngine:
 
     # ------------------------------------------------------------------ #
 
+    def enable_tracing(self, on: bool = True) -> None:
+        """Record per-stage wall times of the GPU pipeline (see
+        ``get_trace``). No-op on the CPU path."""
+        if self._gpu is not None:
+            self._gpu.enable_tracing(on)
+
+    def get_trace(self) -> Optional[dict]:
+        """Stage -> list of ms per shap_values call (None if not tracing)."""
+        return self._gpu.trace if self._gpu is not None else None
+
     def _plan(self, m: int, nsamples: Optional[int]) -> CoalitionPlan:
         key = (m, nsamples)
         if key not in self._plan_cache:

@@ -33,20 +33,27 @@ _TIMING = _os.environ.get("KSHAP_TIMING", "") == "1"
 
 
 class _StageTimer:
-    """Optional per-stage wall timing (KSHAP_TIMING=1): each mark syncs the
-    device, so only use for diagnosis."""
+    """Per-stage wall timing: prints when KSHAP_TIMING=1, records into a
+    sink dict when tracing is enabled (SURVEY.md §5.1 — the reference only
+    had whole-run timeit wall clocks; this gives per-phase numbers). Each
+    mark syncs the device, so enable only for diagnosis."""
 
-    def __init__(self, torch_mod, enabled):
+    def __init__(self, torch_mod, enabled, sink=None):
         self.t = torch_mod
-        self.enabled = enabled
-        self.last = _time.perf_counter() if enabled else 0.0
+        self.enabled = enabled or sink is not None
+        self.sink = sink
+        self.last = _time.perf_counter() if self.enabled else 0.0
 
     def mark(self, name):
         if not self.enabled:
             return
         self.t.cuda.synchronize()
         now = _time.perf_counter()
-        print(f"[kshap-timing] {name}: {(now - self.last) * 1e3:.2f} ms", flush=True)
+        ms = (now - self.last) * 1e3
+        if _TIMING:
+            print(f"[kshap-timing] {name}: {ms:.2f} ms", flush=True)
+        if self.sink is not None:
+            self.sink.setdefault(name, []).append(ms)
         self.last = now
 
 _EPS = 1e-7  # fp32 logit clamp
@@ -105,6 +112,10 @@ class GpuKernelShap:
         # same device blocks avoids caching-allocator churn (sporadic ~90 ms
         # hipMalloc stalls measured at B=2560)
         self._ws: dict = {}
+        self.trace: dict = None  # per-stage ms lists when tracing enabled
+
+    def enable_tracing(self, on: bool = True) -> None:
+        self.trace = {} if on else None
 
     def _buf(self, name, shape, dtype=None):
         t = self.torch
@@ -309,7 +320,7 @@ class GpuKernelShap:
         instance_offset: int = 0,
     ) -> List[np.ndarray]:
         t = self.torch
-        timer = _StageTimer(t, _TIMING)
+        timer = _StageTimer(t, _TIMING, sink=self.trace)
         if t.is_tensor(X):
             b = X.shape[0]
             if X.is_cuda:

@@ -408,3 +408,21 @@ def test_fill_random_masks_wide_m(ext):
     masks2 = torch.zeros_like(masks)
     ext.fill_random_masks(masks2, ne, plan.n_random, cdf, szs, num_paired, 0, ids)
     assert torch.equal(masks, masks2)
+
+
+def test_engine_tracing():
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor, make_adult_like
+
+    data = make_adult_like(n_instances=4, n_background=20, seed=0)
+    pred = LinearPredictor.random(data.X.shape[1], 2, seed=0)
+    eng = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", device="cuda"
+    )
+    eng.enable_tracing()
+    eng.shap_values(data.X)
+    trace = eng.get_trace()
+    for stage in ("varying", "masks", "predict", "wls", "d2h"):
+        assert stage in trace and len(trace[stage]) >= 1
+    eng.enable_tracing(False)
+    assert eng.get_trace() is None
