@@ -132,8 +132,25 @@ class GpuKernelShap:
         t = self.torch
         if self.link_name == "identity":
             return p
-        p = t.clamp(p, _EPS, 1.0 - _EPS)
+        eps = 1e-15 if p.dtype == t.float64 else _EPS
+        p = t.clamp(p, eps, 1.0 - eps)
         return t.log(p / (1.0 - p))
+
+    def _predict_rows_f64(self, rows):
+        """fp64 predict for the per-instance totals: link('logit') amplifies
+        probability saturation (p ~ 1-1e-8 clips to 1-1e-7 in fp32, shifting
+        logit by >2), so fx/fnull go through fp64 — B rows only, negligible."""
+        t = self.torch
+        if self.linear is not None:
+            z = rows.double() @ self.linear["W"].double().T + self.linear["b"].double()
+            a = self.linear["act"]
+            if a == 1:
+                return t.sigmoid(z)
+            if a == 2:
+                return t.softmax(z, dim=-1)
+            return z
+        with t.no_grad():
+            return self.module(rows).double()
 
     def _predict_rows(self, rows):
         """Run the predictor on a device tensor of rows -> (n, n_out) fp32."""
@@ -334,10 +351,11 @@ class GpuKernelShap:
             X_dev = t.tensor(X, dtype=t.float32, device=self.device)
         timer.mark("h2d")
 
-        fx = self._predict_rows(X_dev)                  # (B, n_out)
+        fx = self._predict_rows_f64(X_dev)              # (B, n_out) fp64
         lfx = self._link(fx)
-        lfnull = self._link(self.fnull)
-        total_all = lfx - lfnull[None, :]               # (B, n_out)
+        lfnull64 = self._link(self.fnull.double())
+        total_all = (lfx - lfnull64[None, :]).float()   # (B, n_out)
+        lfnull = lfnull64.float()
 
         phi_full = t.zeros(b, self.n_groups, self.n_out, device=self.device)
 
