@@ -101,17 +101,25 @@ __global__ void fill_random_masks_kernel(
     const int b = blockIdx.x;
     if (b >= B) return;
     const int lane = threadIdx.x & (WAVE - 1);
+    const int wv = threadIdx.x >> 6;     // 8 waves split the row range
+    // fixed per-wave chunking keeps the output deterministic regardless of
+    // launch timing; complement pairs never cross a chunk boundary
+    const int chunk = (n_random + 7) / 8;
+    const int lo = wv * chunk;
+    const int hi = min(lo + chunk, n_random);
+    if (lo >= hi) return;
 
     // size table read via L2/L3 (up to ceil((M-1)/2) <= 128 entries; a
     // fixed-size local copy silently truncates for wide M)
     uint8_t* mrow_base = masks + (size_t)b * S * M;
 
-    int remaining = n_random;
-    int written = 0;
+    int remaining = hi - lo;
+    int written = lo;
     uint32_t iter = 0;
     while (remaining > 0) {
         Philox rng;
-        rng.init(seed, (uint32_t)inst_ids[b], iter, (uint32_t)lane | 0x52000000u);
+        rng.init(seed, (uint32_t)inst_ids[b], iter,
+                 (uint32_t)(wv * WAVE + lane) | 0x52000000u);
         // draw subset size from the residual kernel distribution
         float u = (rng.next_u32() >> 8) * (1.0f / 16777216.0f);
         int si = 0;
@@ -163,22 +171,22 @@ extern "C" void launch_fill_random_masks(
     const int words = (M + 63) / 64;
     switch (words) {
         case 1:
-            fill_random_masks_kernel<1><<<dim3(B), dim3(WAVE), 0, stream>>>(
+            fill_random_masks_kernel<1><<<dim3(B), dim3(8 * WAVE), 0, stream>>>(
                 masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
                 seed, inst_ids);
             break;
         case 2:
-            fill_random_masks_kernel<2><<<dim3(B), dim3(WAVE), 0, stream>>>(
+            fill_random_masks_kernel<2><<<dim3(B), dim3(8 * WAVE), 0, stream>>>(
                 masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
                 seed, inst_ids);
             break;
         case 3:
-            fill_random_masks_kernel<3><<<dim3(B), dim3(WAVE), 0, stream>>>(
+            fill_random_masks_kernel<3><<<dim3(B), dim3(8 * WAVE), 0, stream>>>(
                 masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
                 seed, inst_ids);
             break;
         default:
-            fill_random_masks_kernel<4><<<dim3(B), dim3(WAVE), 0, stream>>>(
+            fill_random_masks_kernel<4><<<dim3(B), dim3(8 * WAVE), 0, stream>>>(
                 masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
                 seed, inst_ids);
             break;
