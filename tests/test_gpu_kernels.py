@@ -43,15 +43,20 @@ def test_fill_random_masks_properties(ext):
     sizes = rnd.sum(axis=2)
     # all rows filled with sizes from the residual range (draw 4-6, compl 6-8)
     assert sizes.min() >= 4 and sizes.max() <= 8
-    # complement pairing: a paired draw (size<=5) is followed by its complement
+    # complement pairing within each of the 8 fixed per-wave chunks (a pair
+    # never crosses a chunk boundary; the last row of a chunk may be an
+    # unpaired truncated draw)
+    chunk = (plan.n_random + 7) // 8
     for bi in range(b):
-        i = 0
-        while i < rnd.shape[1] - 1:
-            if sizes[bi, i] <= 5:
-                assert np.array_equal(rnd[bi, i + 1], 1 - rnd[bi, i])
-                i += 2
-            else:
-                i += 1
+        for clo in range(0, plan.n_random, chunk):
+            chi = min(clo + chunk, plan.n_random)
+            i = clo
+            while i < chi - 1:
+                if sizes[bi, i] <= 5:
+                    assert np.array_equal(rnd[bi, i + 1], 1 - rnd[bi, i])
+                    i += 2
+                else:
+                    i += 1
     # determinism + per-instance keying
     masks2 = torch.zeros_like(masks)
     ext.fill_random_masks(masks2, ne, plan.n_random, cdf, szs, 5, 0, ids)
@@ -396,15 +401,18 @@ def test_fill_random_masks_wide_m(ext):
     sizes = rnd.sum(axis=2)
     lo, hi = plan.random_sizes.min(), plan.random_sizes.max()
     assert sizes.min() >= lo and sizes.max() <= m - lo
-    # paired draws followed by exact complements
+    # paired draws followed by exact complements (within per-wave chunks)
+    chunk = (plan.n_random + 7) // 8
     for bi in range(2):
-        i = 0
-        while i < rnd.shape[1] - 1:
-            if sizes[bi, i] <= num_paired:
-                assert np.array_equal(rnd[bi, i + 1], 1 - rnd[bi, i])
-                i += 2
-            else:
-                i += 1
+        for clo in range(0, plan.n_random, chunk):
+            chi = min(clo + chunk, plan.n_random)
+            i = clo
+            while i < chi - 1:
+                if sizes[bi, i] <= num_paired:
+                    assert np.array_equal(rnd[bi, i + 1], 1 - rnd[bi, i])
+                    i += 2
+                else:
+                    i += 1
     masks2 = torch.zeros_like(masks)
     ext.fill_random_masks(masks2, ne, plan.n_random, cdf, szs, num_paired, 0, ids)
     assert torch.equal(masks, masks2)
