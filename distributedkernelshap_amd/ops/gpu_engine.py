@@ -229,14 +229,12 @@ class GpuKernelShap:
         bgv = self.bg_part[:, vidx]             # (N, m, o)
         return xv.permute(0, 2, 1)[:, :, :, None] - bgv.permute(2, 1, 0)[None]
 
-    def _ey_fused_linear(self, masks, X_dev, varying):
+    def _ey_fused_linear(self, masks, masksT, X_dev, varying):
         """K3-K6 fused MFMA path (linear predictor, Mpad<=64, Npad<=128)."""
         t = self.torch
         b, s, m = masks.shape
-        mpad = max(4, (m + 3) // 4 * 4)
+        mpad = masksT.shape[1]
         npad = (self.N + 15) // 16 * 16
-        masksT = self._buf("masksT", (b, mpad, s))
-        self.ext.transpose_masks(masks, masksT)
         diff = self._buf("diff", (b, self.n_out, mpad, npad))
         diff.zero_()
         diff[:, :, :m, : self.N] = self._diff_tensor(X_dev, varying)
@@ -385,14 +383,32 @@ class GpuKernelShap:
             masks, kw = self._device_masks(plan, gids)
             timer.mark("masks")
             sub_X = X_dev[ids_t]
+            packed = None
+            if m <= 64:
+                # packed u64 masks feed the MFMA WLS Gram build
+                packed = self._buf("packed", (len(ids), plan.nsamples), t.int64)
             if self.linear is not None:
                 mpad = max(4, (m + 3) // 4 * 4)
                 npad = (self.N + 15) // 16 * 16
                 if mpad <= 64 and npad <= 128 and self.n_out in (1, 2, 4):
-                    ey = self._ey_fused_linear(masks, sub_X, varying)
+                    masksT = self._buf("masksT", (len(ids), mpad, plan.nsamples))
+                    self.ext.transpose_masks(masks, masksT, packed)
+                    ey = self._ey_fused_linear(masks, masksT, sub_X, varying)
                 else:
+                    if packed is not None:
+                        self.ext.transpose_masks(
+                            masks,
+                            self._buf("masksT", (len(ids), max(4, (m + 3) // 4 * 4), plan.nsamples)),
+                            packed,
+                        )
                     ey = self._ey_linear_torch(masks, sub_X, varying)
             else:
+                if packed is not None:
+                    self.ext.transpose_masks(
+                        masks,
+                        self._buf("masksT", (len(ids), max(4, (m + 3) // 4 * 4), plan.nsamples)),
+                        packed,
+                    )
                 ey = self._ey_torch_module(masks, sub_X, varying)
             timer.mark("predict")
             # in-place link transform: ey is a workspace, not needed afterwards
@@ -408,7 +424,7 @@ class GpuKernelShap:
             else:
                 phi = self._buf("phi", (len(ids), m, self.n_out))
                 if 2 <= m <= 64 and self.n_out <= 8:
-                    self.ext.wls_solve(masks, kw, ey_adj, total, phi)
+                    self.ext.wls_solve(masks, kw, ey_adj, total, phi, packed)
                 else:
                     phi = self._solve_torch(masks, kw, ey_adj, total)
             timer.mark("wls")

@@ -16,8 +16,8 @@ extern "C" int launch_fused_predict_linear(
     hipStream_t stream);
 
 extern "C" void launch_transpose_masks(
-    const uint8_t* masks, float* masksT, int B, int S, int M, int Mpad,
-    hipStream_t stream);
+    const uint8_t* masks, float* masksT, uint64_t* packed, int B, int S,
+    int M, int Mpad, hipStream_t stream);
 
 extern "C" void launch_synth_chunk(
     const uint8_t* masks, const float* x, const float* bg, const int* col_group,
@@ -25,9 +25,9 @@ extern "C" void launch_synth_chunk(
     int s_hi, hipStream_t stream);
 
 extern "C" int launch_wls_solve(
-    const uint8_t* masks, const float* kw, const float* ey_adj,
-    const float* total, float* phi, int B, int S, int M, int n_out,
-    hipStream_t stream);
+    const uint8_t* masks, const uint64_t* packed, const float* kw,
+    const float* ey_adj, const float* total, float* phi, int B, int S, int M,
+    int n_out, hipStream_t stream);
 
 namespace {
 
@@ -55,15 +55,23 @@ void fill_random_masks(
         current_stream());
 }
 
-void transpose_masks(torch::Tensor masks, torch::Tensor masksT) {
+void transpose_masks(torch::Tensor masks, torch::Tensor masksT,
+                     c10::optional<torch::Tensor> packed) {
     CHECK_DEV(masks); CHECK_DEV(masksT);
     TORCH_CHECK(masks.dtype() == torch::kUInt8 && masksT.dtype() == torch::kFloat32);
     int B = masks.size(0), S = masks.size(1), M = masks.size(2);
     int Mpad = masksT.size(1);
     TORCH_CHECK(masksT.size(0) == B && masksT.size(2) == S && Mpad >= M, "masksT shape");
+    uint64_t* pk = nullptr;
+    if (packed.has_value()) {
+        CHECK_DEV(packed.value());
+        TORCH_CHECK(M <= 64 && packed->size(0) == B && packed->size(1) == S,
+                    "packed shape/M");
+        pk = reinterpret_cast<uint64_t*>(packed->data_ptr<int64_t>());
+    }
     launch_transpose_masks(
-        masks.data_ptr<uint8_t>(), masksT.data_ptr<float>(), B, S, M, Mpad,
-        current_stream());
+        masks.data_ptr<uint8_t>(), masksT.data_ptr<float>(), pk, B, S, M,
+        Mpad, current_stream());
 }
 
 void fused_predict_linear(
@@ -101,13 +109,20 @@ void synth_chunk(
 
 void wls_solve(
     torch::Tensor masks, torch::Tensor kw, torch::Tensor ey_adj,
-    torch::Tensor total, torch::Tensor phi) {
+    torch::Tensor total, torch::Tensor phi,
+    c10::optional<torch::Tensor> packed) {
     CHECK_DEV(masks); CHECK_DEV(kw); CHECK_DEV(ey_adj); CHECK_DEV(total); CHECK_DEV(phi);
     int B = masks.size(0), S = masks.size(1), M = masks.size(2);
     int n_out = ey_adj.size(2);
     TORCH_CHECK(phi.size(0) == B && phi.size(1) == M && phi.size(2) == n_out, "phi shape");
+    const uint64_t* pk = nullptr;
+    if (packed.has_value()) {
+        CHECK_DEV(packed.value());
+        TORCH_CHECK(packed->size(0) == B && packed->size(1) == S, "packed shape");
+        pk = reinterpret_cast<const uint64_t*>(packed->data_ptr<int64_t>());
+    }
     int rc = launch_wls_solve(
-        masks.data_ptr<uint8_t>(), kw.data_ptr<float>(), ey_adj.data_ptr<float>(),
+        masks.data_ptr<uint8_t>(), pk, kw.data_ptr<float>(), ey_adj.data_ptr<float>(),
         total.data_ptr<float>(), phi.data_ptr<float>(), B, S, M, n_out,
         current_stream());
     TORCH_CHECK(rc == 0, "wls_solve: unsupported shape (2<=M<=64, n_out<=8)");
@@ -121,9 +136,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_predict_linear", &fused_predict_linear,
           "MFMA fused mask@diff GEMM + activation + background reduce (K3-K6)");
     m.def("transpose_masks", &transpose_masks,
-          "masks (B,S,M) u8 -> (B,Mpad,S) f32 for coalesced A staging (K2b)");
+          "masks (B,S,M) u8 -> (B,Mpad,S) f32 (+ packed u64) for coalesced "
+          "A staging (K2b)",
+          pybind11::arg("masks"), pybind11::arg("masksT"),
+          pybind11::arg("packed") = pybind11::none());
     m.def("synth_chunk", &synth_chunk,
           "masked-background perturbation synthesis tile (K3')");
     m.def("wls_solve", &wls_solve,
-          "batched constrained WLS Shapley solve (K7)");
+          "batched constrained WLS Shapley solve (K7)",
+          pybind11::arg("masks"), pybind11::arg("kw"), pybind11::arg("ey_adj"),
+          pybind11::arg("total"), pybind11::arg("phi"),
+          pybind11::arg("packed") = pybind11::none());
 }

@@ -209,7 +209,7 @@ extern "C" void launch_fill_random_masks(
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define MAX_MPAD 64        // fused path supports up to 64 varying groups
-#define S_TILE 256         // s-rows per workgroup (4 sub-tiles of 64)
+#define S_TILE 512         // s-rows per workgroup (8 sub-tiles of 64)
 #define S_SUB 64           // rows per sub-tile (4 waves x 16)
 
 template <int NOUT, int ACT>
@@ -313,6 +313,11 @@ void fused_predict_linear_kernel(
             if (ACT == 1) {
 #pragma unroll
                 for (int o = 0; o < NOUT; ++o) z[o] = 1.0f / (1.0f + __expf(-z[o]));
+            } else if (ACT == 2 && NOUT == 2) {
+                // binary softmax = one sigmoid: p1 = 1/(1+exp(z0-z1))
+                float p1 = 1.0f / (1.0f + __expf(z[0] - z[1]));
+                z[0] = 1.0f - p1;
+                z[1] = p1;
             } else if (ACT == 2) {
                 float mx = z[0];
 #pragma unroll
@@ -600,6 +605,7 @@ void wls_solve_kernel(
 __global__ void transpose_masks_kernel(
     const uint8_t* __restrict__ masks,  // (B, S, M)
     float* __restrict__ masksT,         // (B, Mpad, S)
+    uint64_t* __restrict__ packed,      // (B, S) bit i = mask[b,s,i]; or null
     int B, int S, int M, int Mpad)
 {
     const int b = blockIdx.y;
@@ -607,16 +613,23 @@ __global__ void transpose_masks_kernel(
     if (s >= S) return;
     const uint8_t* row = masks + ((size_t)b * S + s) * M;
     float* out = masksT + (size_t)b * Mpad * S + s;
-    for (int k = 0; k < M; ++k) out[(size_t)k * S] = (float)row[k];
+    uint64_t bits = 0ull;
+    for (int k = 0; k < M; ++k) {
+        uint8_t v = row[k] & 1;
+        out[(size_t)k * S] = (float)v;
+        bits |= ((uint64_t)v) << k;
+    }
     for (int k = M; k < Mpad; ++k) out[(size_t)k * S] = 0.0f;
+    if (packed) packed[(size_t)b * S + s] = bits;
 }
 
 extern "C" void launch_transpose_masks(
-    const uint8_t* masks, float* masksT, int B, int S, int M, int Mpad,
-    hipStream_t stream)
+    const uint8_t* masks, float* masksT, uint64_t* packed, int B, int S,
+    int M, int Mpad, hipStream_t stream)
 {
     dim3 grid((S + 255) / 256, B), block(256);
-    transpose_masks_kernel<<<grid, block, 0, stream>>>(masks, masksT, B, S, M, Mpad);
+    transpose_masks_kernel<<<grid, block, 0, stream>>>(
+        masks, masksT, packed, B, S, M, Mpad);
 }
 
 // ------------------------------------------------------------------------- //
@@ -630,7 +643,7 @@ extern "C" void launch_transpose_masks(
 
 __global__ __launch_bounds__(256)
 void wls_solve_mfma_kernel(
-    const uint8_t* __restrict__ masks,   // (B, S, M)
+    const uint64_t* __restrict__ packed, // (B, S) pre-packed mask bits
     const float* __restrict__ kw,        // (B, S)
     const float* __restrict__ ey_adj,    // (B, S, n_out)
     const float* __restrict__ total,     // (B, n_out)
@@ -655,7 +668,7 @@ void wls_solve_mfma_kernel(
 
     if (tid < n_out) tot_s[tid] = total[(size_t)b * n_out + tid];
 
-    const uint8_t* mbase = masks + (size_t)b * S * M;
+    const uint64_t* pbase = packed + (size_t)b * S;
     const float* kwb = kw + (size_t)b * S;
     const float* eyb = ey_adj + (size_t)b * S * n_out;
 
@@ -668,9 +681,7 @@ void wls_solve_mfma_kernel(
         const int clen = min(WLS_CHUNK, S - c0);
         __syncthreads();
         if (tid < clen) {
-            const uint8_t* mrow = mbase + (size_t)(c0 + tid) * M;
-            uint64_t bits = 0ull;
-            for (int g = 0; g < M; ++g) bits |= ((uint64_t)(mrow[g] & 1)) << g;
+            uint64_t bits = pbase[c0 + tid];
             pk[tid] = bits;
             wch[tid] = kwb[c0 + tid];
             float mlast = (float)((bits >> (M - 1)) & 1ull);
@@ -745,16 +756,16 @@ void wls_solve_mfma_kernel(
 }
 
 extern "C" int launch_wls_solve(
-    const uint8_t* masks, const float* kw, const float* ey_adj,
-    const float* total, float* phi, int B, int S, int M, int n_out,
-    hipStream_t stream)
+    const uint8_t* masks, const uint64_t* packed, const float* kw,
+    const float* ey_adj, const float* total, float* phi, int B, int S, int M,
+    int n_out, hipStream_t stream)
 {
     if (M < 2 || M > WLS_MAX_M || n_out > WLS_MAX_NOUT) return -1;
     int mm = M - 1;
     if (mm * (mm + 1) / 2 > 8 * 256) return -1;
-    if (mm + n_out <= 16) {
+    if (mm + n_out <= 16 && packed != nullptr) {
         wls_solve_mfma_kernel<<<dim3(B), dim3(256), 0, stream>>>(
-            masks, kw, ey_adj, total, phi, B, S, M, n_out);
+            packed, kw, ey_adj, total, phi, B, S, M, n_out);
     } else {
         wls_solve_kernel<<<dim3(B), dim3(256), 0, stream>>>(
             masks, kw, ey_adj, total, phi, B, S, M, n_out);

@@ -192,8 +192,16 @@ def test_wls_solve_vs_cpu(ext):
     ey = torch.randn(b, s, n_out, generator=g, device="cuda")
     total = torch.randn(b, n_out, generator=g, device="cuda")
     phi = torch.empty(b, m, n_out, device="cuda")
+    # generic scalar kernel (no packed masks)
     ext.wls_solve(masks, kw, ey, total, phi)
+    # MFMA Gram-build kernel (packed masks; (M-1)+n_out <= 16)
+    masksT = torch.empty(b, 12, s, device="cuda")
+    packed = torch.empty(b, s, dtype=torch.int64, device="cuda")
+    ext.transpose_masks(masks, masksT, packed)
+    phi_mfma = torch.empty_like(phi)
+    ext.wls_solve(masks, kw, ey, total, phi_mfma, packed)
     ph = phi.cpu().numpy()
+    ph_m = phi_mfma.cpu().numpy()
     for i in range(b):
         ref = solve_wls(
             masks[i].cpu().numpy(),
@@ -202,8 +210,10 @@ def test_wls_solve_vs_cpu(ext):
             total[i].double().cpu().numpy(),
         )
         assert np.allclose(ph[i], ref, atol=5e-3, rtol=1e-3), np.abs(ph[i] - ref).max()
+        assert np.allclose(ph_m[i], ref, atol=5e-3, rtol=1e-3), np.abs(ph_m[i] - ref).max()
         # constraint holds exactly by construction
         assert np.allclose(ph[i].sum(axis=0), total[i].cpu().numpy(), atol=1e-4)
+        assert np.allclose(ph_m[i].sum(axis=0), total[i].cpu().numpy(), atol=1e-4)
 
 
 def test_wls_solve_m63(ext):
