@@ -69,8 +69,28 @@ def kernel_shap_postprocess_fn(
 
 
 def _worker_main(worker_id, explainer_type, init_args, init_kwargs, task_q, result_q):
-    """Worker process: build an explainer replica, serve explain/attr requests."""
+    """Worker process: build an explainer replica, serve explain/attr requests.
+
+    CUDA replicas are pinned round-robin to the node's GPUs via
+    HIP_VISIBLE_DEVICES — set before torch is imported in this process
+    (spawn context; torch is only imported lazily by the replica ctor)."""
     try:
+        if init_kwargs.get("device") == "cuda":
+            import os
+
+            n_gpus = int(os.environ.get("KSHAP_POOL_NGPUS", "0"))
+            if not n_gpus:
+                try:
+                    import subprocess
+
+                    out = subprocess.run(
+                        ["rocm-smi", "--showid", "--csv"],
+                        capture_output=True, text=True, timeout=10,
+                    ).stdout
+                    n_gpus = max(1, out.count("card"))
+                except Exception:
+                    n_gpus = 1
+            os.environ["HIP_VISIBLE_DEVICES"] = str(worker_id % n_gpus)
         replica = explainer_type(*init_args, **init_kwargs)
     except Exception as e:  # construction failure must not hang the pool
         result_q.put(("fatal", worker_id, repr(e)))
@@ -195,9 +215,24 @@ class DistributedExplainer:
             self._task_queues[i % self.n_workers].put(
                 ("explain", i, b, base + int(offsets[i]), kwargs)
             )
+        import queue as _queue
+
         unordered: List[Tuple[int, Any]] = []
         while len(unordered) < n_batches:
-            msg = self._result_q.get()
+            try:
+                msg = self._result_q.get(timeout=5)
+            except _queue.Empty:
+                # failure detection (the reference hung
+                # map_unordered forever on a dead actor) — fail fast if any
+                # worker process died
+                dead = [i for i, p in enumerate(self._procs) if not p.is_alive()]
+                if dead:
+                    self.shutdown()
+                    raise RuntimeError(
+                        f"worker process(es) {dead} died; "
+                        f"{len(unordered)}/{n_batches} batches completed"
+                    )
+                continue
             if msg[0] == "result":
                 unordered.append((msg[1], msg[2]))
             elif msg[0] == "error":

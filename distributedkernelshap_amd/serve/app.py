@@ -28,6 +28,37 @@ logger = logging.getLogger(__name__)
 
 __all__ = ["KernelShapModel", "BatchKernelShapModel", "create_app"]
 
+_METRICS = None
+
+
+def _metrics():
+    """Process-wide metric singletons (SURVEY.md §5.5 observability);
+    prometheus registries reject duplicate names, so create_app may run many
+    times but metrics register once."""
+    global _METRICS
+    if _METRICS is None:
+        try:
+            from prometheus_client import (
+                CONTENT_TYPE_LATEST,
+                Counter,
+                Histogram,
+                generate_latest,
+            )
+
+            _METRICS = (
+                Counter("kshap_requests_total", "explain requests"),
+                Histogram("kshap_request_seconds", "explain latency"),
+                Histogram(
+                    "kshap_batch_size", "coalesced batch sizes",
+                    buckets=[1, 2, 4, 8, 16, 32, 64, 128],
+                ),
+                generate_latest,
+                CONTENT_TYPE_LATEST,
+            )
+        except ImportError:  # pragma: no cover
+            _METRICS = (None, None, None, None, None)
+    return _METRICS
+
 
 class KernelShapModel:
     """Serving replica: fit at construction, explain per request
@@ -96,6 +127,8 @@ def create_app(
     app = FastAPI(title="distributedkernelshap-amd")
     queue: asyncio.Queue = asyncio.Queue()
 
+    req_count, req_latency, batch_hist, generate_latest, content_type = _metrics()
+
     async def _batcher():
         while True:
             item = await queue.get()
@@ -111,6 +144,8 @@ def create_app(
                 pass
             payloads = [b[0] for b in batch]
             futures = [b[1] for b in batch]
+            if batch_hist is not None:
+                batch_hist.observe(len(batch))
             loop = asyncio.get_event_loop()
             try:
                 if isinstance(model, BatchKernelShapModel):
@@ -145,12 +180,26 @@ def create_app(
 
     @app.post("/explain")
     async def explain(request: Request):
+        import time
+
+        t0 = time.perf_counter()
         payload = await request.json()
         fut: asyncio.Future = asyncio.get_event_loop().create_future()
         await queue.put((payload, fut))
         body = await fut
+        if req_count is not None:
+            req_count.inc()
+            req_latency.observe(time.perf_counter() - t0)
         from fastapi.responses import Response
 
         return Response(content=body, media_type="application/json")
+
+    @app.get("/metrics")
+    async def metrics():
+        from fastapi.responses import Response
+
+        if req_count is None:
+            return Response(status_code=404)
+        return Response(content=generate_latest(), media_type=content_type)
 
     return app

@@ -166,3 +166,19 @@ def test_sample_sharded_equals_sequential(problem):
         p.join(timeout=60)
     for o in range(2):
         assert np.allclose(sv[o], seq[o], rtol=0, atol=1e-8)
+
+
+def test_pool_dead_worker_fails_fast(problem):
+    """A killed worker must raise, not hang (reference quirk SURVEY.md §2.8 /
+    §5.3: map_unordered blocked forever)."""
+    data, pred = problem
+    ks = KernelShap(
+        pred, link="logit", device="cpu",
+        distributed_opts={"n_workers": 2, "batch_size": 2},
+    )
+    ks.fit(data.background, groups=data.groups)
+    # kill one worker behind the pool's back
+    ks._explainer._procs[0].terminate()
+    ks._explainer._procs[0].join()
+    with pytest.raises(RuntimeError, match="died"):
+        ks._explainer.get_explanation(data.X)

@@ -100,3 +100,16 @@ def test_http_app_concurrent_requests(served):
         assert r.status_code == 200
         sv = np.asarray(r.json()["data"]["shap_values"][0])
         assert sv.shape == (1, 12)
+
+
+def test_metrics_endpoint(served):
+    from fastapi.testclient import TestClient
+
+    data, pred, ckw, fkw = served
+    model = BatchKernelShapModel(pred, data.background, ckw, fkw)
+    app = create_app(model, max_batch_size=4, max_wait_ms=5)
+    with TestClient(app) as client:
+        client.post("/explain", json={"array": data.X[:1].tolist()})
+        r = client.get("/metrics")
+        assert r.status_code == 200
+        assert b"kshap_requests_total" in r.content
