@@ -212,16 +212,18 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define S_TILE 512         // s-rows per workgroup (8 sub-tiles of 64)
 #define S_SUB 64           // rows per sub-tile (4 waves x 16)
 
-template <int NOUT, int ACT>
-__global__ __launch_bounds__(256)
-void fused_predict_linear_kernel(
+template <int NOUT, int ACT, int NT>   // NT = Npad/16 col tiles (constexpr
+__global__ __launch_bounds__(256)      // LDS offsets: the kernel was VALU-
+void fused_predict_linear_kernel(      // bound on runtime address math)
     const float* __restrict__ masksT,   // (B, Mpad, S) zero-padded rows k>=M
-    const float* __restrict__ diff,     // (B, NOUT, Mpad, Npad)
-    const float* __restrict__ base,     // (NOUT, Npad)
-    const float* __restrict__ wbg,      // (Npad)  0 for padding cols
+    const float* __restrict__ diff,     // (B, NOUT, Mpad, NT*16)
+    const float* __restrict__ base,     // (NOUT, NT*16)
+    const float* __restrict__ wbg,      // (NT*16)  0 for padding cols
     float* __restrict__ ey,             // (B, S, NOUT)
-    int B, int S, int M, int Mpad, int Npad)
+    int B, int S, int M, int Mpad)
 {
+    constexpr int NPAD = NT * 16;
+    constexpr int NSTRIDE = NPAD + ((16 - (NPAD & 31)) & 31);  // ≡16 mod 32
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     const int b = blockIdx.x / n_stiles;
     const int stile = blockIdx.x % n_stiles;
@@ -230,133 +232,148 @@ void fused_predict_linear_kernel(
     const int lane = tid & (WAVE - 1);
     const int wave = tid >> 6;           // 0..3
 
-    // LDS: mask tile [Mpad][MSTRIDE], diff [NOUT][Mpad][NSTRIDE], base, wbg
-    const int MSTRIDE = S_SUB + 16;      // 80 ≡ 16 mod 32
+    constexpr int MSTRIDE = S_SUB + 16;  // 80 ≡ 16 mod 32
     extern __shared__ float lds[];
     float* mask_lds = lds;                                   // Mpad*MSTRIDE
-    const int NSTRIDE_PAD = ((16 - (Npad & 31)) & 31);
-    const int NSTRIDE = Npad + NSTRIDE_PAD;                  // ≡16 mod 32
     float* diff_lds = mask_lds + Mpad * MSTRIDE;             // NOUT*Mpad*NSTRIDE
-    float* base_lds = diff_lds + NOUT * Mpad * NSTRIDE;      // NOUT*Npad
-    float* wbg_lds = base_lds + NOUT * Npad;                 // Npad
+    float* base_lds = diff_lds + NOUT * Mpad * NSTRIDE;      // NOUT*NPAD
+    float* wbg_lds = base_lds + NOUT * NPAD;                 // NPAD
 
-    // ---- stage diff / base / wbg once; the 4 s-subtiles reuse them --------
+    // ---- stage diff / base / wbg once; the 8 s-subtiles reuse them --------
     const float* msrc = masksT + (size_t)b * Mpad * S;
-    const float* dsrc = diff + (size_t)b * NOUT * Mpad * Npad;
-    for (int idx = tid; idx < NOUT * Mpad * Npad; idx += 256) {
-        int o = idx / (Mpad * Npad);
-        int rem = idx % (Mpad * Npad);
-        int k = rem / Npad, n = rem % Npad;
-        diff_lds[(o * Mpad + k) * NSTRIDE + n] = dsrc[idx];
+    const float* dsrc = diff + (size_t)b * NOUT * Mpad * NPAD;
+    for (int idx = tid; idx < NOUT * Mpad * NPAD; idx += 256) {
+        int ok = idx / NPAD;             // o * Mpad + k
+        int n = idx % NPAD;
+        diff_lds[ok * NSTRIDE + n] = dsrc[idx];
     }
-    for (int idx = tid; idx < NOUT * Npad; idx += 256) base_lds[idx] = base[idx];
-    for (int idx = tid; idx < Npad; idx += 256) wbg_lds[idx] = wbg[idx];
+    for (int idx = tid; idx < NOUT * NPAD; idx += 256) base_lds[idx] = base[idx];
+    for (int idx = tid; idx < NPAD; idx += 256) wbg_lds[idx] = wbg[idx];
+
+    const int swave = wave * 16;         // this wave's 16 s-rows
+    const int arow = lane & 15;          // A row (s) / B col (n) within tile
+    const int akcol = lane >> 4;         // k within the 4-wide micro-step
+    // per-lane LDS base pointers: all loop offsets are compile-time now
+    const float* dbase = diff_lds + akcol * NSTRIDE + arow;
+    const float* mbase_l = mask_lds + akcol * MSTRIDE + swave + arow;
 
     for (int sub = 0; sub < S_TILE / S_SUB; ++sub) {
-    const int ssub0 = s0 + sub * S_SUB;
-    if (ssub0 >= S) break;
-    // ---- stage mask sub-tile ([k][s], coalesced from transposed layout) ---
-    __syncthreads();   // mask_lds rewritten each sub-tile; diff stable
-    for (int idx = tid; idx < Mpad * S_SUB; idx += 256) {
-        int k = idx / S_SUB, s = idx % S_SUB;
-        mask_lds[k * MSTRIDE + s] = (ssub0 + s < S) ? msrc[(size_t)k * S + ssub0 + s] : 0.0f;
-    }
-    __syncthreads();
+        const int ssub0 = s0 + sub * S_SUB;
+        if (ssub0 >= S) break;
+        // ---- stage mask sub-tile ([k][s], coalesced from transposed src) --
+        __syncthreads();   // mask_lds rewritten each sub-tile; diff stable
+        for (int idx = tid; idx < Mpad * S_SUB; idx += 256) {
+            int k = idx / S_SUB, ss = idx % S_SUB;
+            mask_lds[k * MSTRIDE + ss] =
+                (ssub0 + ss < S) ? msrc[(size_t)k * S + ssub0 + ss] : 0.0f;
+        }
+        __syncthreads();
 
-    // ---- MFMA loop ---------------------------------------------------------
-    const int NT = Npad / 16;            // col tiles (Npad % 16 == 0)
-    const int swave = wave * 16;         // this wave's 16 s-rows
-    const int arow = lane & 15;          // A row (s) within tile
-    const int akcol = lane >> 4;         // A col (k) 0..3
+        f32x4 acc[NT][NOUT];
+#pragma unroll
+        for (int ct = 0; ct < NT; ++ct)
+#pragma unroll
+            for (int o = 0; o < NOUT; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
 
-    // NT <= 8; the ct loop is fully unrolled with a wave-uniform guard so
-    // every acc index is compile-time constant (runtime-indexed ext_vector
-    // arrays are demoted to scratch — guide §5.4 rule 20)
-    f32x4 acc[8][NOUT];
+        for (int ks = 0; ks < Mpad; ks += 4) {
+            float a = mbase_l[ks * MSTRIDE];
 #pragma unroll
-    for (int ct = 0; ct < 8; ++ct)
-#pragma unroll
-        for (int o = 0; o < NOUT; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
-
-    for (int ks = 0; ks < Mpad; ks += 4) {
-        float a = mask_lds[(ks + akcol) * MSTRIDE + swave + arow];
-#pragma unroll
-        for (int ct = 0; ct < 8; ++ct) {
-            if (ct < NT) {
+            for (int ct = 0; ct < NT; ++ct) {
 #pragma unroll
                 for (int o = 0; o < NOUT; ++o) {
-                    float bv = diff_lds[(o * Mpad + ks + akcol) * NSTRIDE + ct * 16 + arow];
-                    acc[ct][o] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[ct][o], 0, 0, 0);
+                    float bv = dbase[(o * Mpad + ks) * NSTRIDE + ct * 16];
+                    acc[ct][o] =
+                        __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[ct][o], 0, 0, 0);
                 }
             }
         }
-    }
 
-    // ---- epilogue: activation + weighted reduction over n ------------------
-    // C/D map: col = lane&15, row = (lane>>4)*4 + reg
-    float partial[NOUT][4];
+        // ---- epilogue: activation + weighted reduction over n -------------
+        // C/D map: col = lane&15, row = (lane>>4)*4 + reg
+        float partial[NOUT][4];
 #pragma unroll
-    for (int o = 0; o < NOUT; ++o)
+        for (int o = 0; o < NOUT; ++o)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) partial[o][r] = 0.0f;
+            for (int r = 0; r < 4; ++r) partial[o][r] = 0.0f;
 
 #pragma unroll
-    for (int ct = 0; ct < 8; ++ct) {
-        if (ct >= NT) break;
-        int n = ct * 16 + (lane & 15);
-        float wn = wbg_lds[n];
+        for (int ct = 0; ct < NT; ++ct) {
+            const int n = ct * 16 + arow;
+            float wn = wbg_lds[n];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            float z[NOUT];
+            for (int r = 0; r < 4; ++r) {
+                float z[NOUT];
 #pragma unroll
-            for (int o = 0; o < NOUT; ++o) z[o] = acc[ct][o][r] + base_lds[o * Npad + n];
-            if (ACT == 1) {
+                for (int o = 0; o < NOUT; ++o) z[o] = acc[ct][o][r] + base_lds[o * NPAD + n];
+                if (ACT == 1) {
 #pragma unroll
-                for (int o = 0; o < NOUT; ++o) z[o] = 1.0f / (1.0f + __expf(-z[o]));
-            } else if (ACT == 2 && NOUT == 2) {
-                // binary softmax = one sigmoid: p1 = 1/(1+exp(z0-z1))
-                float p1 = 1.0f / (1.0f + __expf(z[0] - z[1]));
-                z[0] = 1.0f - p1;
-                z[1] = p1;
-            } else if (ACT == 2) {
-                float mx = z[0];
+                    for (int o = 0; o < NOUT; ++o) z[o] = 1.0f / (1.0f + __expf(-z[o]));
+                } else if (ACT == 2 && NOUT == 2) {
+                    // binary softmax = one sigmoid: p1 = 1/(1+exp(z0-z1))
+                    float p1 = 1.0f / (1.0f + __expf(z[0] - z[1]));
+                    z[0] = 1.0f - p1;
+                    z[1] = p1;
+                } else if (ACT == 2) {
+                    float mx = z[0];
 #pragma unroll
-                for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
-                float sum = 0.0f;
+                    for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
+                    float sum = 0.0f;
 #pragma unroll
-                for (int o = 0; o < NOUT; ++o) { z[o] = __expf(z[o] - mx); sum += z[o]; }
-                float inv = 1.0f / sum;
+                    for (int o = 0; o < NOUT; ++o) { z[o] = __expf(z[o] - mx); sum += z[o]; }
+                    float inv = 1.0f / sum;
 #pragma unroll
-                for (int o = 0; o < NOUT; ++o) z[o] *= inv;
-            }
+                    for (int o = 0; o < NOUT; ++o) z[o] *= inv;
+                }
 #pragma unroll
-            for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * z[o];
-        }
-    }
-    // reduce over the 16 lanes of each row group (xor bits 0-3 stay in-group)
-#pragma unroll
-    for (int o = 0; o < NOUT; ++o)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            float v = partial[o][r];
-            v += __shfl_xor(v, 1);
-            v += __shfl_xor(v, 2);
-            v += __shfl_xor(v, 4);
-            v += __shfl_xor(v, 8);
-            partial[o][r] = v;
-        }
-    if ((lane & 15) == 0) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            int s = ssub0 + swave + (lane >> 4) * 4 + r;
-            if (s < S) {
-#pragma unroll
-                for (int o = 0; o < NOUT; ++o)
-                    ey[((size_t)b * S + s) * NOUT + o] = partial[o][r];
+                for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * z[o];
             }
         }
-    }
+        // reduce over the 16 lanes of each row group (xor bits 0-3 in-group)
+#pragma unroll
+        for (int o = 0; o < NOUT; ++o)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float v = partial[o][r];
+                v += __shfl_xor(v, 1);
+                v += __shfl_xor(v, 2);
+                v += __shfl_xor(v, 4);
+                v += __shfl_xor(v, 8);
+                partial[o][r] = v;
+            }
+        if (arow == 0) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int ss = ssub0 + swave + akcol * 4 + r;
+                if (ss < S) {
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o)
+                        ey[((size_t)b * S + ss) * NOUT + o] = partial[o][r];
+                }
+            }
+        }
     }  // sub-tile loop
+}
+
+template <int NOUT, int ACT>
+static void launch_fused_nt(
+    const float* masksT, const float* diff, const float* base, const float* wbg,
+    float* ey, int B, int S, int M, int Mpad, int Npad, hipStream_t stream)
+{
+    const int n_stiles = (S + S_TILE - 1) / S_TILE;
+    dim3 grid(B * n_stiles), block(256);
+    const int NSTRIDE = Npad + ((16 - (Npad & 31)) & 31);
+    size_t lds = (size_t)(Mpad * (S_SUB + 16) + NOUT * Mpad * NSTRIDE
+                          + NOUT * Npad + Npad) * 4;
+#define KSHAP_CASE(NTV) \
+    case NTV: \
+        fused_predict_linear_kernel<NOUT, ACT, NTV><<<grid, block, lds, stream>>>( \
+            masksT, diff, base, wbg, ey, B, S, M, Mpad); \
+        break;
+    switch (Npad / 16) {
+        KSHAP_CASE(1) KSHAP_CASE(2) KSHAP_CASE(3) KSHAP_CASE(4)
+        KSHAP_CASE(5) KSHAP_CASE(6) KSHAP_CASE(7) KSHAP_CASE(8)
+    }
+#undef KSHAP_CASE
 }
 
 template <int NOUT>
@@ -364,23 +381,15 @@ static void launch_fused_act(
     const float* masksT, const float* diff, const float* base, const float* wbg,
     float* ey, int B, int S, int M, int Mpad, int Npad, int act, hipStream_t stream)
 {
-    const int n_stiles = (S + S_TILE - 1) / S_TILE;
-    dim3 grid(B * n_stiles), block(256);
-    const int MSTRIDE = S_SUB + 16;
-    const int NSTRIDE = Npad + ((16 - (Npad & 31)) & 31);
-    size_t lds = (size_t)(Mpad * MSTRIDE + NOUT * Mpad * NSTRIDE + NOUT * Npad + Npad) * 4;
     switch (act) {
         case 0:
-            fused_predict_linear_kernel<NOUT, 0><<<grid, block, lds, stream>>>(
-                masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad);
+            launch_fused_nt<NOUT, 0>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
         case 1:
-            fused_predict_linear_kernel<NOUT, 1><<<grid, block, lds, stream>>>(
-                masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad);
+            launch_fused_nt<NOUT, 1>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
         default:
-            fused_predict_linear_kernel<NOUT, 2><<<grid, block, lds, stream>>>(
-                masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad);
+            launch_fused_nt<NOUT, 2>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
     }
 }
