@@ -118,6 +118,11 @@ class GpuKernelShap:
         # hipMalloc stalls measured at B=2560)
         self._ws: dict = {}
         self.trace: dict = None  # per-stage ms lists when tracing enabled
+        # hipGraph capture of the fused pipeline (~60 fixed-shape launches):
+        # keyed by (B, nsamples, varying pattern, offset); disable with
+        # KSHAP_GRAPH=0 or automatically on capture failure
+        self._graphs: dict = {}
+        self._graphs_enabled = _os.environ.get("KSHAP_GRAPH", "1") == "1"
 
     def enable_tracing(self, on: bool = True) -> None:
         self.trace = {} if on else None
@@ -125,10 +130,11 @@ class GpuKernelShap:
     def _buf(self, name, shape, dtype=None):
         t = self.torch
         dtype = dtype or t.float32
-        buf = self._ws.get(name)
-        if buf is None or buf.shape != tuple(shape) or buf.dtype != dtype:
+        key = (name, tuple(shape), dtype)   # per-shape: captured hipGraphs
+        buf = self._ws.get(key)             # hold raw pointers to these
+        if buf is None:
             buf = t.empty(*shape, dtype=dtype, device=self.device)
-            self._ws[name] = buf
+            self._ws[key] = buf
         return buf
 
     # ------------------------------------------------------------------ #
@@ -225,16 +231,17 @@ class GpuKernelShap:
 
     # ------------------------------------------------------------------ #
 
-    def _diff_tensor(self, X_dev, varying):
+    def _diff_tensor(self, X_dev, varying, vidx_t=None):
         """diff[b, o, k, n] = x_part[b, k, o] - bg_part[n, k, o] over varying
         groups k (the algebraically folded masked-background blend)."""
         t = self.torch
-        vidx = t.tensor(varying, dtype=t.int64, device=self.device)
+        vidx = (vidx_t if vidx_t is not None
+                else t.tensor(varying, dtype=t.int64, device=self.device))
         xv = self._x_part(X_dev)[:, vidx]       # (b, m, o)
         bgv = self.bg_part[:, vidx]             # (N, m, o)
         return xv.permute(0, 2, 1)[:, :, :, None] - bgv.permute(2, 1, 0)[None]
 
-    def _ey_fused_linear(self, masks, masksT, X_dev, varying):
+    def _ey_fused_linear(self, masks, masksT, X_dev, varying, vidx_t=None):
         """K3-K6 fused MFMA path (linear predictor, Mpad<=64, Npad<=128)."""
         t = self.torch
         b, s, m = masks.shape
@@ -242,7 +249,7 @@ class GpuKernelShap:
         npad = (self.N + 15) // 16 * 16
         diff = self._buf("diff", (b, self.n_out, mpad, npad))
         diff.zero_()
-        diff[:, :, :m, : self.N] = self._diff_tensor(X_dev, varying)
+        diff[:, :, :m, : self.N] = self._diff_tensor(X_dev, varying, vidx_t)
         base = t.zeros(self.n_out, npad, device=self.device)
         base[:, : self.N] = self.baseN.T
         wbg = t.zeros(npad, device=self.device)
@@ -331,6 +338,79 @@ class GpuKernelShap:
         return ey
 
     # ------------------------------------------------------------------ #
+    # hipGraph fast path
+
+    def _fused_body(self, X_dev, plan, varying, vidx_t, instance_offset, phi_full):
+        """The captureable single-bucket fused pipeline: X_dev -> phi_full.
+        Pure device ops (no pageable H2D, no host syncs)."""
+        t = self.torch
+        b = X_dev.shape[0]
+        m = len(varying)
+        fx = self._predict_rows_f64(X_dev)
+        lfx = self._link(fx)
+        lfnull64 = self._link(self.fnull.double())
+        total = (lfx - lfnull64[None, :]).float().contiguous()
+        lfnull = lfnull64.float()
+        gids = np.arange(b) + instance_offset
+        masks, kw = self._device_masks(plan, gids)
+        mpad = max(4, (m + 3) // 4 * 4)
+        packed = self._buf("packed", (b, plan.nsamples), t.int64)
+        masksT = self._buf("masksT", (b, mpad, plan.nsamples))
+        self.ext.transpose_masks(masks, masksT, packed)
+        ey = self._ey_fused_linear(masks, masksT, X_dev, varying, vidx_t)
+        if self.link_name == "identity":
+            ey_adj = ey.sub_(lfnull[None, None, :])
+        else:
+            ey.clamp_(_EPS, 1.0 - _EPS)
+            ey.log_().sub_(t.log1p(-t.exp(ey)))
+            ey_adj = ey.sub_(lfnull[None, None, :])
+        phi = self._buf("phi", (b, m, self.n_out))
+        self.ext.wls_solve(masks, kw, ey_adj, total, phi, packed)
+        phi_full.zero_()
+        phi_full[:, vidx_t] = phi
+
+    def _graph_explain(self, X_dev, plan, varying, instance_offset):
+        """Capture-or-replay the fused pipeline; returns shap values or None
+        when graphs are unavailable."""
+        t = self.torch
+        b = X_dev.shape[0]
+        key = (b, plan.nsamples, varying.tobytes(), int(instance_offset))
+        entry = self._graphs.get(key)
+        if entry is None:
+            try:
+                vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
+                x_static = self._buf("Xg", (b, self.D))
+                phi_full = t.zeros(
+                    b, self.n_groups, self.n_out, device=self.device
+                )
+                x_static.copy_(X_dev)
+                # warmup on a side stream (required before capture)
+                side = t.cuda.Stream()
+                side.wait_stream(t.cuda.current_stream())
+                with t.cuda.stream(side):
+                    self._fused_body(
+                        x_static, plan, varying, vidx_t, instance_offset, phi_full
+                    )
+                t.cuda.current_stream().wait_stream(side)
+                graph = t.cuda.CUDAGraph()
+                with t.cuda.graph(graph):
+                    self._fused_body(
+                        x_static, plan, varying, vidx_t, instance_offset, phi_full
+                    )
+                entry = (graph, x_static, phi_full, vidx_t)
+                self._graphs[key] = entry
+            except Exception as e:  # pragma: no cover - capture support varies
+                logger.warning(
+                    "hipGraph capture failed (%r); continuing eagerly", e
+                )
+                self._graphs_enabled = False
+                return None
+        graph, x_static, phi_full, _ = entry
+        if x_static.data_ptr() != X_dev.data_ptr():
+            x_static.copy_(X_dev)
+        graph.replay()
+        out = phi_full.double().cpu().numpy()
+        return [np.ascontiguousarray(out[:, :, o]) for o in range(self.n_out)]
 
     def shap_values(
         self,
@@ -372,6 +452,24 @@ class GpuKernelShap:
         else:
             uniq, inverse = np.unique(vmat, axis=0, return_inverse=True)
         timer.mark("bucket")
+
+        if self._graphs_enabled and uniq.shape[0] == 1 and self.linear is not None:
+            varying0 = np.nonzero(uniq[0])[0]
+            m0 = len(varying0)
+            if m0 >= 2:
+                plan0 = self.engine._plan(m0, nsamples)
+                mpad0 = max(4, (m0 + 3) // 4 * 4)
+                npad0 = (self.N + 15) // 16 * 16
+                if (
+                    mpad0 <= 64
+                    and npad0 <= 128
+                    and self.n_out in (1, 2, 4)
+                    and not self._l1_active(plan0, l1_reg)
+                ):
+                    out = self._graph_explain(X_dev, plan0, varying0, instance_offset)
+                    if out is not None:
+                        timer.mark("graph")
+                        return out
 
         for u in range(uniq.shape[0]):
             varying = np.nonzero(uniq[u])[0]
