@@ -40,28 +40,31 @@ def _server_main(host, port, max_batch_size, assets_dir, device):
     uvicorn.run(app, host=host, port=port, log_level="warning")
 
 
-def distribute_requests(X, url, max_workers):
-    """Fan out one GET-style request per instance (reference
+def distribute_requests(X, urls, max_workers):
+    """Fan out one request per instance, round-robin over replicas (reference
     ``distribute_request``/``explain``, serve_explanations.py:96-139)."""
     import httpx
 
     instances = np.split(X, X.shape[0])
 
-    def post(x):
+    def post(item):
+        i, x = item
         with httpx.Client() as client:
-            r = client.post(url, json={"array": x.tolist()}, timeout=120.0)
+            r = client.post(
+                urls[i % len(urls)], json={"array": x.tolist()}, timeout=120.0
+            )
             r.raise_for_status()
             return r.text
 
     with concurrent.futures.ThreadPoolExecutor(max_workers=max_workers) as pool:
-        return list(pool.map(post, instances))
+        return list(pool.map(post, enumerate(instances)))
 
 
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--replicas", type=int, default=1,
-                        help="kept for filename parity; one GPU replica "
-                             "saturates the node")
+                        help="server replica processes (round-robin clients); "
+                             "one GPU replica already saturates a device")
     parser.add_argument("--max-batch-size", type=int, default=64)
     parser.add_argument("--instances", type=int, default=2560)
     parser.add_argument("--nruns", type=int, default=3)
@@ -80,26 +83,33 @@ def main():
     assert X.shape[0] == args.instances
 
     ctx = mp.get_context("spawn")
-    server = ctx.Process(
-        target=_server_main,
-        args=(args.host, args.port, args.max_batch_size, args.assets_dir,
-              args.device),
-        daemon=True,
-    )
-    server.start()
-    url = f"http://{args.host}:{args.port}/explain"
-    # wait for readiness
+    servers = []
+    urls = []
+    for r in range(args.replicas):
+        port = args.port + r
+        srv = ctx.Process(
+            target=_server_main,
+            args=(args.host, port, args.max_batch_size, args.assets_dir,
+                  args.device),
+            daemon=True,
+        )
+        srv.start()
+        servers.append(srv)
+        urls.append(f"http://{args.host}:{port}/explain")
+    # wait for readiness of every replica
     import httpx
 
-    for _ in range(600):
-        try:
-            if httpx.get(f"http://{args.host}:{args.port}/healthz",
-                         timeout=2.0).status_code == 200:
-                break
-        except Exception:
-            time.sleep(0.5)
-    else:
-        raise RuntimeError("server did not become ready")
+    for r in range(args.replicas):
+        for _ in range(600):
+            try:
+                if httpx.get(
+                    f"http://{args.host}:{args.port + r}/healthz", timeout=2.0
+                ).status_code == 200:
+                    break
+            except Exception:
+                time.sleep(0.5)
+        else:
+            raise RuntimeError(f"server replica {r} did not become ready")
 
     path = get_filename(
         args.replicas, 0, serve=True, max_batch_size=args.max_batch_size,
@@ -110,7 +120,7 @@ def main():
         for run in range(args.nruns):
             logger.info("run %d/%d", run + 1, args.nruns)
             t_start = timeit.default_timer()
-            responses = distribute_requests(X, url, args.concurrency)
+            responses = distribute_requests(X, urls, args.concurrency)
             t_elapsed = timeit.default_timer() - t_start
             logger.info("Time elapsed: %.4f s (%d responses)",
                         t_elapsed, len(responses))
@@ -120,8 +130,10 @@ def main():
             with open(path, "wb") as f:
                 pickle.dump(result, f)
     finally:
-        server.terminate()
-        server.join(timeout=10)
+        for srv in servers:
+            srv.terminate()
+        for srv in servers:
+            srv.join(timeout=10)
 
 
 if __name__ == "__main__":
