@@ -81,6 +81,16 @@ def main():
     data = load_data(args.assets_dir)
     X = data.X_test[: args.instances]
     assert X.shape[0] == args.instances
+    if not os.path.exists(os.path.join(args.assets_dir, "predictor.pkl")):
+        logger.info("no fitted model found; running scripts/fit_model.py")
+        import subprocess
+
+        subprocess.run(
+            [sys.executable,
+             os.path.join(os.path.dirname(__file__), "..", "scripts", "fit_model.py"),
+             "--assets-dir", args.assets_dir],
+            check=True,
+        )
 
     ctx = mp.get_context("spawn")
     servers = []
