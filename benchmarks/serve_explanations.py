@@ -43,18 +43,23 @@ def _server_main(host, port, max_batch_size, assets_dir, device):
 def distribute_requests(X, urls, max_workers):
     """Fan out one request per instance, round-robin over replicas (reference
     ``distribute_request``/``explain``, serve_explanations.py:96-139)."""
+    import threading
+
     import httpx
 
     instances = np.split(X, X.shape[0])
+    tls = threading.local()
 
     def post(item):
         i, x = item
-        with httpx.Client() as client:
-            r = client.post(
-                urls[i % len(urls)], json={"array": x.tolist()}, timeout=120.0
-            )
-            r.raise_for_status()
-            return r.text
+        client = getattr(tls, "client", None)
+        if client is None:
+            client = tls.client = httpx.Client()
+        r = client.post(
+            urls[i % len(urls)], json={"array": x.tolist()}, timeout=120.0
+        )
+        r.raise_for_status()
+        return r.text
 
     with concurrent.futures.ThreadPoolExecutor(max_workers=max_workers) as pool:
         return list(pool.map(post, enumerate(instances)))

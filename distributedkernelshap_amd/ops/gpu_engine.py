@@ -122,7 +122,11 @@ class GpuKernelShap:
         # keyed by (B, nsamples, varying pattern, offset); disable with
         # KSHAP_GRAPH=0 or automatically on capture failure
         self._graphs: dict = {}
+        self._graph_hits: dict = {}
         self._graphs_enabled = _os.environ.get("KSHAP_GRAPH", "1") == "1"
+        # capture costs ~100 ms: only worth it for recurring shapes (a
+        # serving workload with varying batch sizes must stay eager)
+        self._graph_min_hits = int(_os.environ.get("KSHAP_GRAPH_MIN_HITS", "2"))
 
     def enable_tracing(self, on: bool = True) -> None:
         self.trace = {} if on else None
@@ -377,6 +381,10 @@ class GpuKernelShap:
         key = (b, plan.nsamples, varying.tobytes(), int(instance_offset))
         entry = self._graphs.get(key)
         if entry is None:
+            hits = self._graph_hits.get(key, 0) + 1
+            self._graph_hits[key] = hits
+            if hits < self._graph_min_hits:
+                return None        # eager until the shape proves recurring
             try:
                 vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
                 x_static = self._buf("Xg", (b, self.D))
