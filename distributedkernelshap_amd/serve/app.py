@@ -147,6 +147,11 @@ def create_app(
             if batch_hist is not None:
                 batch_hist.observe(len(batch))
             loop = asyncio.get_event_loop()
+            import os as _os
+            import time as _time
+
+            _dbg = _os.environ.get("KSHAP_TIMING") == "1"
+            _t0 = _time.perf_counter() if _dbg else 0.0
             try:
                 if isinstance(model, BatchKernelShapModel):
                     results = await loop.run_in_executor(None, model.batch, payloads)
@@ -154,6 +159,12 @@ def create_app(
                     results = []
                     for p in payloads:
                         results.append(await loop.run_in_executor(None, model, p))
+                if _dbg:
+                    print(
+                        f"[kshap-serve] batch={len(batch)} "
+                        f"explain_ms={(_time.perf_counter() - _t0) * 1e3:.1f}",
+                        flush=True,
+                    )
                 for fut, res in zip(futures, results):
                     if not fut.done():
                         fut.set_result(res)
