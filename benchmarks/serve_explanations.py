@@ -40,14 +40,21 @@ def _server_main(host, port, max_batch_size, assets_dir, device):
     uvicorn.run(app, host=host, port=port, log_level="warning")
 
 
-def distribute_requests(X, urls, max_workers):
-    """Fan out one request per instance, round-robin over replicas (reference
-    ``distribute_request``/``explain``, serve_explanations.py:96-139)."""
+def distribute_requests(X, urls, max_workers, batch_mode="ray",
+                        max_batch_size=64):
+    """Fan requests out over replicas. 'ray' mode sends one request per
+    instance (reference ``distribute_request``/``explain``,
+    serve_explanations.py:96-139); 'default' pre-splits into client-side
+    minibatches (reference k8s variant batch_mode)."""
     import threading
 
     import httpx
 
-    instances = np.split(X, X.shape[0])
+    if batch_mode == "default":
+        n_batches = (X.shape[0] + max_batch_size - 1) // max_batch_size
+        instances = np.array_split(X, n_batches)
+    else:
+        instances = np.split(X, X.shape[0])
     tls = threading.local()
 
     def post(item):
@@ -73,7 +80,12 @@ def main():
     parser.add_argument("--max-batch-size", type=int, default=64)
     parser.add_argument("--instances", type=int, default=2560)
     parser.add_argument("--nruns", type=int, default=3)
-    parser.add_argument("--concurrency", type=int, default=64)
+    parser.add_argument("--concurrency", type=int, default=32)
+    parser.add_argument("--batch-mode", default="ray", choices=["ray", "default"],
+                        help="'ray': one request per instance (server-side "
+                             "coalescing); 'default': client-side minibatches "
+                             "of --max-batch-size (reference "
+                             "k8s_serve_explanations.py:180-185 parity)")
     parser.add_argument("--host", default="127.0.0.1")
     parser.add_argument("--port", type=int, default=8800)
     parser.add_argument("--device", default="auto")
@@ -135,7 +147,9 @@ def main():
         for run in range(args.nruns):
             logger.info("run %d/%d", run + 1, args.nruns)
             t_start = timeit.default_timer()
-            responses = distribute_requests(X, urls, args.concurrency)
+            responses = distribute_requests(
+                X, urls, args.concurrency, args.batch_mode, args.max_batch_size
+            )
             t_elapsed = timeit.default_timer() - t_start
             logger.info("Time elapsed: %.4f s (%d responses)",
                         t_elapsed, len(responses))
