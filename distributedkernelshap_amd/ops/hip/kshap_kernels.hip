@@ -232,10 +232,8 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
     const int lane = tid & (WAVE - 1);
     const int wave = tid >> 6;           // 0..3
 
-    constexpr int MSTRIDE = S_SUB + 16;  // 80 ≡ 16 mod 32
     extern __shared__ float lds[];
-    float* mask_lds = lds;                                   // Mpad*MSTRIDE
-    float* diff_lds = mask_lds + Mpad * MSTRIDE;             // NOUT*Mpad*NSTRIDE
+    float* diff_lds = lds;                                   // NOUT*Mpad*NSTRIDE
     float* base_lds = diff_lds + NOUT * Mpad * NSTRIDE;      // NOUT*NPAD
     float* wbg_lds = base_lds + NOUT * NPAD;                 // NPAD
 
@@ -249,25 +247,24 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
     }
     for (int idx = tid; idx < NOUT * NPAD; idx += 256) base_lds[idx] = base[idx];
     for (int idx = tid; idx < NPAD; idx += 256) wbg_lds[idx] = wbg[idx];
+    __syncthreads();
 
     const int swave = wave * 16;         // this wave's 16 s-rows
     const int arow = lane & 15;          // A row (s) / B col (n) within tile
     const int akcol = lane >> 4;         // k within the 4-wide micro-step
-    // per-lane LDS base pointers: all loop offsets are compile-time now
+    // per-lane LDS base pointer: all loop offsets are compile-time now
     const float* dbase = diff_lds + akcol * NSTRIDE + arow;
-    const float* mbase_l = mask_lds + akcol * MSTRIDE + swave + arow;
+    // A-operand straight from the transposed global layout: every mask
+    // element is consumed exactly ONCE per launch, so LDS staging (and its
+    // per-sub-tile barriers) was pure overhead; the 16-lane groups read 64 B
+    // consecutive segments (coalesced stream)
+    const float* mlane = msrc + (size_t)akcol * S + swave + arow;
 
     for (int sub = 0; sub < S_TILE / S_SUB; ++sub) {
         const int ssub0 = s0 + sub * S_SUB;
         if (ssub0 >= S) break;
-        // ---- stage mask sub-tile ([k][s], coalesced from transposed src) --
-        __syncthreads();   // mask_lds rewritten each sub-tile; diff stable
-        for (int idx = tid; idx < Mpad * S_SUB; idx += 256) {
-            int k = idx / S_SUB, ss = idx % S_SUB;
-            mask_lds[k * MSTRIDE + ss] =
-                (ssub0 + ss < S) ? msrc[(size_t)k * S + ssub0 + ss] : 0.0f;
-        }
-        __syncthreads();
+        const int srow = ssub0 + swave + arow;
+        const bool svalid = srow < S;
 
         f32x4 acc[NT][NOUT];
 #pragma unroll
@@ -276,7 +273,7 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
             for (int o = 0; o < NOUT; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
 
         for (int ks = 0; ks < Mpad; ks += 4) {
-            float a = mbase_l[ks * MSTRIDE];
+            float a = svalid ? mlane[(size_t)ks * S + ssub0] : 0.0f;
 #pragma unroll
             for (int ct = 0; ct < NT; ++ct) {
 #pragma unroll
@@ -362,8 +359,7 @@ static void launch_fused_nt(
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     dim3 grid(B * n_stiles), block(256);
     const int NSTRIDE = Npad + ((16 - (Npad & 31)) & 31);
-    size_t lds = (size_t)(Mpad * (S_SUB + 16) + NOUT * Mpad * NSTRIDE
-                          + NOUT * Npad + Npad) * 4;
+    size_t lds = (size_t)(NOUT * Mpad * NSTRIDE + NOUT * Npad + Npad) * 4;
 #define KSHAP_CASE(NTV) \
     case NTV: \
         fused_predict_linear_kernel<NOUT, ACT, NTV><<<grid, block, lds, stream>>>( \
