@@ -217,6 +217,11 @@ class KernelShap(Explainer, FitMixin):
         self.distribute = False
         self.distributed_opts = dict(DISTRIBUTED_OPTS)
         if distributed_opts is not None:
+            distributed_opts = dict(distributed_opts)
+            # reference spelling (explainers/kernel_shap.py:210-214) used
+            # 'n_cpus' for the worker count; accept it as an alias
+            if "n_cpus" in distributed_opts and "n_workers" not in distributed_opts:
+                distributed_opts["n_workers"] = distributed_opts.pop("n_cpus")
             self.distributed_opts.update(distributed_opts)
             workers = self.distributed_opts.get("n_workers")
             if workers is not None and workers > 1:
