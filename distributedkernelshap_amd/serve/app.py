@@ -189,7 +189,7 @@ def create_app(
     async def expected_value():
         return {"expected_value": np.asarray(model.explainer.expected_value).tolist()}
 
-    @app.post("/explain")
+    @app.api_route("/explain", methods=["GET", "POST"])
     async def explain(request: Request):
         import time
 
