@@ -102,6 +102,23 @@ class KernelShapEngine:
         self.D = d
         self.N = n
         self._plan_cache: dict = {}
+        # algebraic fast path for linear predictors (same folding as the GPU
+        # kernel): logits(synth[s,n]) = base[n] + sum_k mask[s,k]*diff[k,n]
+        self._lin = None
+        lp = getattr(predictor, "linear_params", None)
+        if lp is not None:
+            W, bias, act = lp()
+            W = np.asarray(W, dtype=np.float64)
+            contrib = self.background[:, :, None] * W.T[None, :, :]  # (N,D,o)
+            bg_part = np.zeros((n, self.n_groups, W.shape[0]))
+            np.add.at(bg_part, (slice(None), self._col_group), contrib)
+            self._lin = {
+                "W": W,
+                "b": np.asarray(bias, dtype=np.float64),
+                "act": act,
+                "bg_part": bg_part,
+                "base": self.background @ W.T + np.asarray(bias),
+            }
         self._gpu = None
         if self.device == "cuda":
             from ..ops import gpu_engine
@@ -210,10 +227,12 @@ class KernelShapEngine:
     def _ey(self, x: np.ndarray, masks: np.ndarray, varying: np.ndarray) -> np.ndarray:
         """K3-K5: masked-background synth -> predict -> weighted mean over bg.
 
-        CPU reference path; chunked so the synth matrix never exceeds
-        ``chunk_rows`` rows (the GPU path fuses these stages instead —
-        SURVEY.md §7.1).
+        Linear predictors take the algebraically folded path (no synth
+        materialisation); arbitrary callables use the chunked synth path
+        (the CI oracle for the GPU synth kernel).
         """
+        if self._lin is not None:
+            return self._ey_linear(x, masks, varying)
         s = masks.shape[0]
         n, d = self.background.shape
         # expand group mask (S, m_varying) to column mask (S, D)
@@ -232,6 +251,37 @@ class KernelShapEngine:
             y = y.reshape(hi - lo, n, self.n_out)
             ey[lo:hi] = np.einsum("cno,n->co", y, self.bg_weights)
         return ey
+
+    def _ey_linear(self, x: np.ndarray, masks: np.ndarray, varying: np.ndarray) -> np.ndarray:
+        lin = self._lin
+        w = lin["W"]
+        x_contrib = x[:, None] * w.T                       # (D, o)
+        x_part = np.zeros((self.n_groups, w.shape[0]))
+        np.add.at(x_part, self._col_group, x_contrib)
+        diff = x_part[None, varying] - lin["bg_part"][:, varying]   # (N, m, o)
+        act = lin["act"]
+        mf = masks.astype(np.float64)
+        if act == "softmax" and diff.shape[2] == 2:
+            # binary softmax = one sigmoid on the logit difference: a single
+            # (S,m)@(m,N) BLAS GEMM + one exp pass
+            dd = diff[:, :, 1] - diff[:, :, 0]              # (N, m)
+            based = lin["base"][:, 1] - lin["base"][:, 0]   # (N,)
+            zd = mf @ dd.T + based[None, :]
+            p1 = 1.0 / (1.0 + np.exp(-zd))
+            ey1 = p1 @ self.bg_weights
+            return np.stack([1.0 - ey1, ey1], axis=1)
+        # logits[s, n, o] = base[n, o] + masks[s] . diff[n]
+        zz = np.einsum("sm,nmo->sno", mf, diff)
+        zz += lin["base"][None, :, :]
+        if act == "sigmoid":
+            p = 1.0 / (1.0 + np.exp(-zz))
+        elif act == "softmax":
+            zz -= zz.max(axis=-1, keepdims=True)
+            np.exp(zz, out=zz)
+            p = zz / zz.sum(axis=-1, keepdims=True)
+        else:
+            p = zz
+        return np.einsum("sno,n->so", p, self.bg_weights)
 
     # ------------------------------------------------------------------ #
 

@@ -187,23 +187,27 @@ def sample_masks(
     if plan.n_random > 0:
         rng = np.random.Generator(np.random.Philox(key=[seed, instance_index]))
         num_paired = int(np.floor((m - 1) / 2.0))
-        pos = ne
-        end = plan.nsamples
-        while pos < end:
-            size = int(
-                rng.choice(plan.random_sizes, p=plan.random_size_probs)
-            )
-            perm = rng.permutation(m)[:size]
-            masks[pos, :] = 0
-            masks[pos, perm] = 1
-            weights[pos] = 1.0
-            pos += 1
-            # complement pair (shap adds the complement when the size is paired)
-            if pos < end and size <= num_paired:
-                masks[pos] = 1 - masks[pos - 1]
-                weights[pos] = 1.0
-                pos += 1
-        rnd = weights[ne:]
-        weights[ne:] = rnd * (plan.weight_left / rnd.sum())
+        n_rand = plan.n_random
+        # vectorised random phase: draw all subset sizes at once, allocate
+        # rows by prefix sum (paired draws consume 2 rows), build subsets via
+        # the rank-threshold trick
+        sizes_draw = rng.choice(
+            plan.random_sizes, size=n_rand, p=plan.random_size_probs
+        )
+        paired = sizes_draw <= num_paired
+        rows_per = 1 + paired.astype(np.int64)
+        cum = np.cumsum(rows_per)
+        jcut = int(np.searchsorted(cum, n_rand))  # first draw reaching budget
+        sizes_draw = sizes_draw[: jcut + 1]
+        paired = paired[: jcut + 1]
+        starts = cum[: jcut + 1] - rows_per[: jcut + 1]
+        # random subset of size s: the s smallest ranks of a random row
+        rand = rng.random((jcut + 1, m))
+        ranks = rand.argsort(axis=1).argsort(axis=1)
+        draw_masks = (ranks < sizes_draw[:, None]).astype(np.uint8)
+        masks[ne + starts] = draw_masks
+        comp_ok = paired & (starts + 1 < n_rand)
+        masks[ne + starts[comp_ok] + 1] = 1 - draw_masks[comp_ok]
+        weights[ne:] = plan.weight_left / n_rand
 
     return masks, weights
