@@ -67,7 +67,21 @@ def main():
     parser.add_argument("--device", default="auto")
     parser.add_argument("--assets-dir", default="assets")
     parser.add_argument("--results-dir", default="results")
+    parser.add_argument("--config-file", default=None,
+                        help="TOML config (distributedkernelshap_amd.config) "
+                             "overriding bench/engine defaults")
     args = parser.parse_args()
+    if args.config_file:
+        from distributedkernelshap_amd.config import Config
+
+        cfg = Config.from_toml(args.config_file)
+        args.instances = cfg.bench.instances
+        args.nruns = cfg.bench.nruns
+        args.batch = cfg.bench.batch_sizes
+        args.results_dir = cfg.bench.results_dir
+        args.assets_dir = cfg.bench.assets_dir
+        if cfg.distributed.n_workers:
+            args.workers = cfg.distributed.n_workers
 
     data = load_data(args.assets_dir)
     model_path = os.path.join(args.assets_dir, "predictor.pkl")

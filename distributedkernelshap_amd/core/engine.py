@@ -64,7 +64,11 @@ class KernelShapEngine:
         seed: int = 0,
         device: str = "auto",
         chunk_rows: int = 1 << 20,
+        kernels=None,
     ):
+        from ..config import KernelConfig
+
+        self.kernels = kernels if kernels is not None else KernelConfig()
         self.predictor = predictor
         self.background = np.ascontiguousarray(_as_2d(background), dtype=np.float64)
         n, d = self.background.shape

@@ -287,6 +287,8 @@ class GpuKernelShap:
         return ey
 
     def _ey_torch_module(self, masks, X_dev, varying, chunk_rows=None):
+        if chunk_rows is None and self.engine.kernels.synth_chunk_rows:
+            chunk_rows = max(self.N, self.engine.kernels.synth_chunk_rows)
         """K3' synth + torch predictor + weighted mean (arbitrary-predictor
         path). Synth tiles never leave the device (SURVEY.md §7.3); multiple
         instances are packed per predictor call so the python/launch overhead
@@ -461,7 +463,9 @@ class GpuKernelShap:
             uniq, inverse = np.unique(vmat, axis=0, return_inverse=True)
         timer.mark("bucket")
 
-        if self._graphs_enabled and uniq.shape[0] == 1 and self.linear is not None:
+        kc = self.engine.kernels
+        if self._graphs_enabled and uniq.shape[0] == 1 and self.linear is not None \
+                and kc.fused_predict and kc.wls_mode in ("auto", "mfma"):
             varying0 = np.nonzero(uniq[0])[0]
             m0 = len(varying0)
             if m0 >= 2:
@@ -501,7 +505,8 @@ class GpuKernelShap:
             if self.linear is not None:
                 mpad = max(4, (m + 3) // 4 * 4)
                 npad = (self.N + 15) // 16 * 16
-                if mpad <= 64 and npad <= 128 and self.n_out in (1, 2, 4):
+                if (kc.fused_predict and mpad <= 64 and npad <= 128
+                        and self.n_out in (1, 2, 4)):
                     masksT = self._buf("masksT", (len(ids), mpad, plan.nsamples))
                     self.ext.transpose_masks(masks, masksT, packed)
                     ey = self._ey_fused_linear(masks, masksT, sub_X, varying)
@@ -534,10 +539,13 @@ class GpuKernelShap:
                 phi = self._solve_host_l1(masks, kw, ey_adj, total, l1_reg)
             else:
                 phi = self._buf("phi", (len(ids), m, self.n_out))
-                if 2 <= m <= 64 and self.n_out <= 8:
-                    self.ext.wls_solve(masks, kw, ey_adj, total, phi, packed)
-                else:
+                if kc.wls_mode == "torch" or not (2 <= m <= 64 and self.n_out <= 8):
                     phi = self._solve_torch(masks, kw, ey_adj, total)
+                else:
+                    # wls_mode 'generic' disables the MFMA Gram build by
+                    # withholding the packed masks
+                    pk = None if kc.wls_mode == "generic" else packed
+                    self.ext.wls_solve(masks, kw, ey_adj, total, phi, pk)
             timer.mark("wls")
             vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
             phi_full[ids_t[:, None], vidx_t[None, :]] = phi

@@ -112,3 +112,32 @@ def test_load_data_and_model_cache(tmp_path):
     ref = clf.predict_proba(data.X_test[:50])
     assert np.allclose(probs, ref, atol=1e-10)
     assert hasattr(pred, "linear_params")
+
+
+def test_pool_cli_config_file(tmp_path):
+    import subprocess
+    import sys
+
+    cfgf = tmp_path / "c.toml"
+    cfgf.write_text(
+        f"""
+[bench]
+instances = 6
+nruns = 1
+batch_sizes = [3]
+results_dir = "{tmp_path}/res"
+assets_dir = "{tmp_path}/assets"
+
+[distributed]
+n_workers = 2
+"""
+    )
+    root = os.path.join(os.path.dirname(__file__), "..")
+    r = subprocess.run(
+        [sys.executable, "benchmarks/pool.py", "--config-file", str(cfgf)],
+        cwd=root, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-500:]
+    assert os.path.exists(
+        os.path.join(str(tmp_path), "res", "ray_workers_2_bsize_3_actorfr_1.0.pkl")
+    )
