@@ -16,13 +16,12 @@ Pipeline per instance:
 from __future__ import annotations
 
 import logging
-from math import comb
 from typing import Callable, List, Optional, Sequence, Union
 
 import numpy as np
 
 from .links import convert_to_link
-from .sampler import CoalitionPlan, default_nsamples, plan_coalitions, sample_masks
+from .sampler import CoalitionPlan, plan_coalitions, sample_masks
 from .solver import solve_wls
 
 logger = logging.getLogger(__name__)

@@ -24,7 +24,7 @@ from __future__ import annotations
 
 import logging
 import multiprocessing as mp
-from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
+from typing import Any, Callable, Dict, List, Sequence, Tuple
 
 import numpy as np
 

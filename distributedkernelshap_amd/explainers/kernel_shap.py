@@ -11,7 +11,6 @@ from __future__ import annotations
 
 import logging
 import warnings
-from functools import singledispatch
 from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple, Union
 
 import numpy as np
@@ -508,10 +507,6 @@ class KernelShap(Explainer, FitMixin):
                 summarised = True
 
         raw_pred = np.asarray(self.predictor(X))
-        from ..core.links import convert_to_link
-
-        linkf, _ = convert_to_link(self.link)
-        link_pred = linkf(raw_pred)
         if raw_pred.ndim > 1 and raw_pred.shape[1] > 1:
             prediction = np.argmax(raw_pred, axis=1)
         else:
