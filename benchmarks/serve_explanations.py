@@ -22,7 +22,12 @@ logging.basicConfig(level=logging.INFO)
 logger = logging.getLogger(__name__)
 
 
-def _server_main(host, port, max_batch_size, assets_dir, device):
+def _server_main(host, port, max_batch_size, assets_dir, device, replica_id=0):
+    # pin each replica to a GPU round-robin (before torch import)
+    if device in ("auto", "cuda"):
+        n = os.environ.get("KSHAP_POOL_NGPUS")
+        if n and int(n) > 0:
+            os.environ["HIP_VISIBLE_DEVICES"] = str(replica_id % int(n))
     import uvicorn
 
     from distributedkernelshap_amd.serve import BatchKernelShapModel, create_app
@@ -117,7 +122,7 @@ def main():
         srv = ctx.Process(
             target=_server_main,
             args=(args.host, port, args.max_batch_size, args.assets_dir,
-                  args.device),
+                  args.device, r),
             daemon=True,
         )
         srv.start()

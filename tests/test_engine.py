@@ -141,3 +141,30 @@ def test_l1_reg_num_features(adult_like, linear_predictor):
     sv = eng.shap_values(adult_like.X[:2], l1_reg="num_features(5)")
     # at most 5 nonzero features per instance (plus constraint back-substitution)
     assert (np.abs(sv[0]) > 1e-12).sum(axis=1).max() <= 6
+
+
+def test_single_instance_and_1d_input(adult_like, linear_predictor):
+    eng = KernelShapEngine(
+        linear_predictor, adult_like.background, groups=adult_like.groups,
+        link="logit", device="cpu",
+    )
+    sv = eng.shap_values(adult_like.X[0])  # 1-D input
+    assert sv[0].shape == (1, 12)
+
+
+def test_linear_fast_path_matches_general(rng):
+    """The algebraic linear CPU path must agree with the generic synth path
+    (same predictor wrapped as a plain callable loses linear_params)."""
+    d = 8
+    w = rng.normal(size=(2, d))
+    b = rng.normal(size=2)
+    pred = LinearPredictor(w, b, activation="softmax")
+    plain = lambda X: pred(X)  # noqa: E731  (drops linear_params)
+    bg = rng.normal(size=(20, d))
+    X = rng.normal(size=(3, d))
+    fast = KernelShapEngine(pred, bg, link="logit", seed=0, device="cpu")
+    slow = KernelShapEngine(plain, bg, link="logit", seed=0, device="cpu")
+    svf = fast.shap_values(X)
+    svs = slow.shap_values(X)
+    for o in range(2):
+        assert np.allclose(svf[o], svs[o], atol=1e-10)
