@@ -287,8 +287,6 @@ class GpuKernelShap:
         return ey
 
     def _ey_torch_module(self, masks, X_dev, varying, chunk_rows=None):
-        if chunk_rows is None and self.engine.kernels.synth_chunk_rows:
-            chunk_rows = max(self.N, self.engine.kernels.synth_chunk_rows)
         """K3' synth + torch predictor + weighted mean (arbitrary-predictor
         path). Synth tiles never leave the device (SURVEY.md §7.3); multiple
         instances are packed per predictor call so the python/launch overhead
@@ -296,9 +294,11 @@ class GpuKernelShap:
         t = self.torch
         b, s, m = masks.shape
         if chunk_rows is None:
-            chunk_rows = max(1 << 19, self.N)
-            # cap the synth buffer at ~2 GB
-            chunk_rows = min(chunk_rows, max(self.N, (1 << 29) // max(1, self.D)))
+            chunk_rows = max(self.N, self.engine.kernels.synth_chunk_rows or 1 << 19)
+        # hard cap: the synth buffer must stay ~2 GB regardless of config
+        chunk_rows = min(chunk_rows, max(self.N, (1 << 29) // max(1, self.D)))
+        # never allocate more than the whole batch needs
+        chunk_rows = min(chunk_rows, b * s * self.N)
         ey = t.empty(b, s, self.n_out, device=self.device)
         # masks cover varying groups only; map non-varying columns to a
         # sentinel always-zero mask column m
@@ -314,7 +314,7 @@ class GpuKernelShap:
         rows_per_inst = s * self.N
         if rows_per_inst <= chunk_rows:
             # pack g instances per call
-            g_inst = max(1, chunk_rows // rows_per_inst)
+            g_inst = max(1, min(b, chunk_rows // rows_per_inst))
             buf = self._buf("synth", (g_inst * rows_per_inst, self.D))
             for lo in range(0, b, g_inst):
                 hi = min(lo + g_inst, b)
