@@ -124,7 +124,8 @@ def main():
             results_dir=args.results_dir,
         )
         run_explainer(explainer, X, batch_size, args.nruns, path)
-        explainer._explainer.shutdown()
+        if hasattr(explainer._explainer, "shutdown"):
+            explainer._explainer.shutdown()
 
 
 if __name__ == "__main__":

@@ -223,7 +223,9 @@ class KernelShap(Explainer, FitMixin):
                 distributed_opts["n_workers"] = distributed_opts.pop("n_cpus")
             self.distributed_opts.update(distributed_opts)
             workers = self.distributed_opts.get("n_workers")
-            if workers is not None and workers > 1:
+            # n_workers=1 is a pool of one replica (reference parity);
+            # only None / 0 / -1 mean in-process sequential
+            if workers is not None and workers >= 1:
                 self.distribute = True
         self.distributed_opts["algorithm"] = "kernel_shap"
         self._update_metadata(
