@@ -182,3 +182,44 @@ def test_pool_dead_worker_fails_fast(problem):
     ks._explainer._procs[0].join()
     with pytest.raises(RuntimeError, match="died"):
         ks._explainer.get_explanation(data.X)
+
+
+def test_explain_checkpointed_resume(tmp_path, problem):
+    """Interrupted checkpointed run resumes without recomputing and matches
+    the uninterrupted result exactly."""
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.parallel.checkpointed import (
+        explain_checkpointed,
+    )
+
+    data, pred = problem
+    eng = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cpu",
+    )
+    full = eng.shap_values(data.X)
+    out = str(tmp_path / "ckpt")
+    # simulate interruption: run only the first chunk, then 'crash'
+    calls = {"n": 0}
+    orig = eng.shap_values
+
+    def crashing(X, **kw):
+        calls["n"] += 1
+        if calls["n"] > 1:
+            raise KeyboardInterrupt
+        return orig(X, **kw)
+
+    eng.shap_values = crashing
+    with pytest.raises(KeyboardInterrupt):
+        explain_checkpointed(eng, data.X, out, chunk_instances=3)
+    eng.shap_values = orig
+    # resume: only the remaining chunks run
+    before = calls["n"]
+    sv = explain_checkpointed(eng, data.X, out, chunk_instances=3)
+    for o in range(2):
+        assert np.allclose(sv[o], full[o], rtol=0, atol=1e-10)
+    # first chunk was not recomputed
+    import json as _json
+
+    with open(out + "/manifest.json") as f:
+        assert len(_json.load(f)["done"]) == 3

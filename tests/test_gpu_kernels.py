@@ -480,3 +480,37 @@ def test_graph_replay_tracks_new_inputs():
     assert np.abs(total - fx[:, 0]).max() < 1e-3
     # replay result must differ from X1's (sanity that inputs propagated)
     assert not np.allclose(r2[0], r1[0])
+
+
+def test_engine_gpu_mixed_varying_buckets():
+    """Instances with different varying-group patterns exercise the
+    multi-bucket GPU path (m==0, m==1 and two m>=2 buckets) and must match
+    the CPU oracle."""
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor
+
+    rng = np.random.Generator(np.random.Philox(key=[21, 3]))
+    d = 9
+    pred = LinearPredictor.random(d, 2, seed=3)
+    bg = rng.normal(size=(30, d))
+    mu = bg.mean(axis=0)
+    X = rng.normal(size=(6, d))
+    X[1] = bg[0]              # m == 0 for the exact-background row? no: differs
+    X[1, :] = bg[0, :]        # matches ONE row but differs from others -> varies
+    X[2, 3:] = bg[:, 3:].mean(axis=0)  # still varies (mean != each row)
+    # construct true non-varying columns: constant background columns
+    bg[:, 0] = 1.5
+    bg[:, 1] = -2.0
+    X[3, 0] = 1.5             # group 0 fixed for instance 3
+    X[4, 0] = 1.5
+    X[4, 1] = -2.0            # groups 0,1 fixed for instance 4
+    cpu = KernelShapEngine(pred, bg, link="logit", seed=0, device="cpu")
+    gpu = KernelShapEngine(pred, bg, link="logit", seed=0, device="cuda")
+    sv_c = cpu.shap_values(X)
+    sv_g = gpu.shap_values(X)
+    for o in range(2):
+        assert np.allclose(sv_g[o], sv_c[o], atol=5e-4, rtol=1e-3), np.abs(
+            sv_g[o] - sv_c[o]
+        ).max()
+    # instance 4's fixed groups get exactly zero attribution
+    assert sv_g[0][4, 0] == 0.0 and sv_g[0][4, 1] == 0.0
