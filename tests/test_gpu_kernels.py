@@ -514,3 +514,19 @@ def test_engine_gpu_mixed_varying_buckets():
         ).max()
     # instance 4's fixed groups get exactly zero attribution
     assert sv_g[0][4, 0] == 0.0 and sv_g[0][4, 1] == 0.0
+
+
+def test_engine_gpu_l1_path():
+    """Explicit l1_reg routes through the host LARS path on GPU inputs."""
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor, make_adult_like
+
+    data = make_adult_like(n_instances=4, n_background=30, seed=6)
+    pred = LinearPredictor.random(data.X.shape[1], 2, seed=6)
+    eng = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cuda",
+    )
+    sv = eng.shap_values(data.X, l1_reg="num_features(5)")
+    assert sv[0].shape == (4, 12)
+    assert (np.abs(sv[0]) > 1e-12).sum(axis=1).max() <= 6
