@@ -539,13 +539,23 @@ class GpuKernelShap:
                 phi = self._solve_host_l1(masks, kw, ey_adj, total, l1_reg)
             else:
                 phi = self._buf("phi", (len(ids), m, self.n_out))
-                if kc.wls_mode == "torch" or not (2 <= m <= 64 and self.n_out <= 8):
-                    phi = self._solve_torch(masks, kw, ey_adj, total)
-                else:
+                # fp32 Gram conditioning degrades with M (Shapley kernel
+                # weights span orders of magnitude): MFMA path covers
+                # (M-1)+n_out <= 16; the scalar fp32 kernel up to M=24;
+                # larger M goes to the fp64 batched torch solve
+                use_kernel = (
+                    kc.wls_mode != "torch"
+                    and 2 <= m
+                    and self.n_out <= 8
+                    and (m - 1 + self.n_out <= 16 or m <= 24)
+                )
+                if use_kernel:
                     # wls_mode 'generic' disables the MFMA Gram build by
                     # withholding the packed masks
                     pk = None if kc.wls_mode == "generic" else packed
                     self.ext.wls_solve(masks, kw, ey_adj, total, phi, pk)
+                else:
+                    phi = self._solve_torch(masks, kw, ey_adj, total)
             timer.mark("wls")
             vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
             phi_full[ids_t[:, None], vidx_t[None, :]] = phi
