@@ -530,3 +530,25 @@ def test_engine_gpu_l1_path():
     sv = eng.shap_values(data.X, l1_reg="num_features(5)")
     assert sv[0].shape == (4, 12)
     assert (np.abs(sv[0]) > 1e-12).sum(axis=1).max() <= 6
+
+
+def test_gpu_pipeline_bitwise_deterministic():
+    """Two identical runs produce bitwise-identical shap values — the
+    share-nothing/stream-ordered design has no racy accumulation (SURVEY.md
+    §5.2 rebuild requirement)."""
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor, make_adult_like
+
+    data = make_adult_like(n_instances=64, n_background=100, seed=8)
+    pred = LinearPredictor.random(data.X.shape[1], 2, seed=8)
+
+    def run():
+        eng = KernelShapEngine(
+            pred, data.background, groups=data.groups, link="logit", seed=0,
+            device="cuda",
+        )
+        return eng.shap_values(data.X)
+
+    a, b = run(), run()
+    for o in range(2):
+        assert np.array_equal(a[o], b[o])
