@@ -422,6 +422,18 @@ class GpuKernelShap:
         out = phi_full.double().cpu().numpy()
         return [np.ascontiguousarray(out[:, :, o]) for o in range(self.n_out)]
 
+    # device-memory budget for per-call mask/workspace tensors; batches
+    # whose projected footprint exceeds it are processed in instance chunks
+    _CHUNK_BYTES = 2 << 30
+
+    def _instances_per_chunk(self, nsamples: Optional[int]) -> int:
+        from ..core.sampler import default_nsamples
+
+        s = nsamples or default_nsamples(max(2, self.n_groups))
+        # masks u8 + masksT f32 + packed u64 + kwb f32 + ey f32 per instance
+        per_inst = s * (5 * self.n_groups + 8 + 4 + 4 * self.n_out) + (1 << 12)
+        return max(1, self._CHUNK_BYTES // per_inst)
+
     def shap_values(
         self,
         X: np.ndarray,
@@ -430,6 +442,22 @@ class GpuKernelShap:
         instance_offset: int = 0,
     ) -> List[np.ndarray]:
         t = self.torch
+        b_total = X.shape[0]
+        chunk = self._instances_per_chunk(nsamples)
+        if b_total > chunk:
+            # large batches stream through in fixed-size chunks (1M-instance
+            # configs: the whole-batch mask tensor would be TB-scale)
+            parts = [
+                self.shap_values(
+                    X[lo : lo + chunk], nsamples=nsamples, l1_reg=l1_reg,
+                    instance_offset=instance_offset + lo,
+                )
+                for lo in range(0, b_total, chunk)
+            ]
+            return [
+                np.concatenate([p[o] for p in parts], axis=0)
+                for o in range(self.n_out)
+            ]
         timer = _StageTimer(t, _TIMING, sink=self.trace)
         if t.is_tensor(X):
             b = X.shape[0]
