@@ -194,8 +194,10 @@ class GpuKernelShap:
 
     # ------------------------------------------------------------------ #
 
-    def _device_masks(self, plan: CoalitionPlan, inst_ids: np.ndarray):
-        """masks (B,S,M) u8 + kernel weights (B,S) f32 on device."""
+    def _device_masks(self, plan: CoalitionPlan, inst_ids: np.ndarray,
+                      ids_dev=None):
+        """masks (B,S,M) u8 + kernel weights (B,S) f32 on device.
+        ``ids_dev`` overrides the RNG-key tensor (hipGraph static input)."""
         t = self.torch
         b = len(inst_ids)
         s, m = plan.nsamples, plan.m
@@ -218,12 +220,16 @@ class GpuKernelShap:
         kw[:ne] = ew
         if plan.n_random > 0:
             num_paired = int(np.floor((m - 1) / 2.0))
-            base0 = int(inst_ids[0])
-            if np.array_equal(inst_ids, np.arange(base0, base0 + b)):
-                # contiguous ids (the common bucket): build on device, no H2D
-                ids = t.arange(base0, base0 + b, dtype=t.int32, device=self.device)
+            if ids_dev is not None:
+                ids = ids_dev
             else:
-                ids = t.tensor(inst_ids.astype(np.int32), device=self.device)
+                base0 = int(inst_ids[0])
+                if np.array_equal(inst_ids, np.arange(base0, base0 + b)):
+                    # contiguous ids (common bucket): device arange, no H2D
+                    ids = t.arange(base0, base0 + b, dtype=t.int32,
+                                   device=self.device)
+                else:
+                    ids = t.tensor(inst_ids.astype(np.int32), device=self.device)
             self.ext.fill_random_masks(
                 masks, ne, plan.n_random, cdf, szs, num_paired,
                 int(self.engine.seed), ids,
@@ -346,9 +352,10 @@ class GpuKernelShap:
     # ------------------------------------------------------------------ #
     # hipGraph fast path
 
-    def _fused_body(self, X_dev, plan, varying, vidx_t, instance_offset, phi_full):
+    def _fused_body(self, X_dev, plan, varying, vidx_t, ids_dev, phi_full):
         """The captureable single-bucket fused pipeline: X_dev -> phi_full.
-        Pure device ops (no pageable H2D, no host syncs)."""
+        Pure device ops (no pageable H2D, no host syncs); instance RNG keys
+        come from the static ``ids_dev`` buffer."""
         t = self.torch
         b = X_dev.shape[0]
         m = len(varying)
@@ -357,8 +364,7 @@ class GpuKernelShap:
         lfnull64 = self._link(self.fnull.double())
         total = (lfx - lfnull64[None, :]).float().contiguous()
         lfnull = lfnull64.float()
-        gids = np.arange(b) + instance_offset
-        masks, kw = self._device_masks(plan, gids)
+        masks, kw = self._device_masks(plan, np.arange(b), ids_dev=ids_dev)
         mpad = max(4, (m + 3) // 4 * 4)
         packed = self._buf("packed", (b, plan.nsamples), t.int64)
         masksT = self._buf("masksT", (b, mpad, plan.nsamples))
@@ -380,34 +386,41 @@ class GpuKernelShap:
         when graphs are unavailable."""
         t = self.torch
         b = X_dev.shape[0]
-        key = (b, plan.nsamples, varying.tobytes(), int(instance_offset))
+        # instance ids are a graph INPUT (static buffer), so the key is
+        # offset-independent: chunked 1M-instance sweeps replay one graph
+        key = (b, plan.nsamples, varying.tobytes())
         entry = self._graphs.get(key)
         if entry is None:
             hits = self._graph_hits.get(key, 0) + 1
             self._graph_hits[key] = hits
-            if hits < self._graph_min_hits:
+            if hits < self._graph_min_hits or len(self._graphs) >= 8:
                 return None        # eager until the shape proves recurring
             try:
                 vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
                 x_static = self._buf("Xg", (b, self.D))
+                ids_static = self._buf("idsg", (b,), t.int32)
                 phi_full = t.zeros(
                     b, self.n_groups, self.n_out, device=self.device
                 )
                 x_static.copy_(X_dev)
+                ids_static.copy_(
+                    t.arange(instance_offset, instance_offset + b,
+                             dtype=t.int32, device=self.device)
+                )
                 # warmup on a side stream (required before capture)
                 side = t.cuda.Stream()
                 side.wait_stream(t.cuda.current_stream())
                 with t.cuda.stream(side):
                     self._fused_body(
-                        x_static, plan, varying, vidx_t, instance_offset, phi_full
+                        x_static, plan, varying, vidx_t, ids_static, phi_full
                     )
                 t.cuda.current_stream().wait_stream(side)
                 graph = t.cuda.CUDAGraph()
                 with t.cuda.graph(graph):
                     self._fused_body(
-                        x_static, plan, varying, vidx_t, instance_offset, phi_full
+                        x_static, plan, varying, vidx_t, ids_static, phi_full
                     )
-                entry = (graph, x_static, phi_full, vidx_t)
+                entry = (graph, x_static, ids_static, phi_full, vidx_t)
                 self._graphs[key] = entry
             except Exception as e:  # pragma: no cover - capture support varies
                 logger.warning(
@@ -415,9 +428,13 @@ class GpuKernelShap:
                 )
                 self._graphs_enabled = False
                 return None
-        graph, x_static, phi_full, _ = entry
+        graph, x_static, ids_static, phi_full, _ = entry
         if x_static.data_ptr() != X_dev.data_ptr():
             x_static.copy_(X_dev)
+        ids_static.copy_(
+            t.arange(instance_offset, instance_offset + b, dtype=t.int32,
+                     device=self.device)
+        )
         graph.replay()
         out = phi_full.double().cpu().numpy()
         return [np.ascontiguousarray(out[:, :, o]) for o in range(self.n_out)]
