@@ -266,27 +266,10 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
         const int srow = ssub0 + swave + arow;
         const bool svalid = srow < S;
 
-        f32x4 acc[NT][NOUT];
-#pragma unroll
-        for (int ct = 0; ct < NT; ++ct)
-#pragma unroll
-            for (int o = 0; o < NOUT; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
-
-        for (int ks = 0; ks < Mpad; ks += 4) {
-            float a = svalid ? mlane[(size_t)ks * S + ssub0] : 0.0f;
-#pragma unroll
-            for (int ct = 0; ct < NT; ++ct) {
-#pragma unroll
-                for (int o = 0; o < NOUT; ++o) {
-                    float bv = dbase[(o * Mpad + ks) * NSTRIDE + ct * 16];
-                    acc[ct][o] =
-                        __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[ct][o], 0, 0, 0);
-                }
-            }
-        }
-
-        // ---- epilogue: activation + weighted reduction over n -------------
-        // C/D map: col = lane&15, row = (lane>>4)*4 + reg
+        // column tiles processed in two halves: halves re-read the (cheap,
+        // L2-resident) A stream but halve the accumulator AGPR footprint,
+        // buying occupancy (93 VGPR + 56 AGPR -> occ 3 at NT=7 otherwise)
+        constexpr int NTH = (NT + 1) / 2;
         float partial[NOUT][4];
 #pragma unroll
         for (int o = 0; o < NOUT; ++o)
@@ -294,8 +277,38 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
             for (int r = 0; r < 4; ++r) partial[o][r] = 0.0f;
 
 #pragma unroll
-        for (int ct = 0; ct < NT; ++ct) {
-            const int n = ct * 16 + arow;
+        for (int half = 0; half < 2; ++half) {
+        const int CT0 = half * NTH;
+        const int CTN = half == 0 ? NTH : NT - NTH;
+        if (CTN <= 0) continue;
+
+        f32x4 acc[NTH][NOUT];
+#pragma unroll
+        for (int ct = 0; ct < NTH; ++ct)
+#pragma unroll
+            for (int o = 0; o < NOUT; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
+
+        for (int ks = 0; ks < Mpad; ks += 4) {
+            float a = svalid ? mlane[(size_t)ks * S + ssub0] : 0.0f;
+#pragma unroll
+            for (int ct = 0; ct < NTH; ++ct) {
+                if (ct < CTN) {
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) {
+                        float bv = dbase[(o * Mpad + ks) * NSTRIDE + (CT0 + ct) * 16];
+                        acc[ct][o] =
+                            __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[ct][o], 0, 0, 0);
+                    }
+                }
+            }
+        }
+
+        // ---- epilogue: activation + weighted reduction over n -------------
+        // C/D map: col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+        for (int ct = 0; ct < NTH; ++ct) {
+            if (ct >= CTN) break;
+            const int n = (CT0 + ct) * 16 + arow;
             float wn = wbg_lds[n];
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
@@ -325,6 +338,7 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
                 for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * z[o];
             }
         }
+        }  // half loop
         // reduce over the 16 lanes of each row group (xor bits 0-3 in-group)
 #pragma unroll
         for (int o = 0; o < NOUT; ++o)
