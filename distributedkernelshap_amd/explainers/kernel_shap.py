@@ -509,7 +509,9 @@ class KernelShap(Explainer, FitMixin):
                 summarised = True
 
         raw_pred = np.asarray(self.predictor(X))
-        if raw_pred.ndim > 1 and raw_pred.shape[1] > 1:
+        if self.task == "regression":
+            prediction = raw_pred.reshape(raw_pred.shape[0], -1)
+        elif raw_pred.ndim > 1 and raw_pred.shape[1] > 1:
             prediction = np.argmax(raw_pred, axis=1)
         else:
             prediction = (raw_pred > 0.5).astype(int).reshape(-1)
