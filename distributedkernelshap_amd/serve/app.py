@@ -20,6 +20,8 @@ Native replacement for the reference's ray-serve deployment
 import asyncio
 import json
 import logging
+import os
+import time
 from typing import Any, Dict, List, Optional
 
 import numpy as np
@@ -147,11 +149,8 @@ def create_app(
             if batch_hist is not None:
                 batch_hist.observe(len(batch))
             loop = asyncio.get_event_loop()
-            import os as _os
-            import time as _time
-
-            _dbg = _os.environ.get("KSHAP_TIMING") == "1"
-            _t0 = _time.perf_counter() if _dbg else 0.0
+            _dbg = os.environ.get("KSHAP_TIMING") == "1"
+            _t0 = time.perf_counter() if _dbg else 0.0
             try:
                 if isinstance(model, BatchKernelShapModel):
                     results = await loop.run_in_executor(None, model.batch, payloads)
@@ -162,7 +161,7 @@ def create_app(
                 if _dbg:
                     print(
                         f"[kshap-serve] batch={len(batch)} "
-                        f"explain_ms={(_time.perf_counter() - _t0) * 1e3:.1f}",
+                        f"explain_ms={(time.perf_counter() - _t0) * 1e3:.1f}",
                         flush=True,
                     )
                 for fut, res in zip(futures, results):
@@ -191,8 +190,6 @@ def create_app(
 
     @app.api_route("/explain", methods=["GET", "POST"])
     async def explain(request: Request):
-        import time
-
         t0 = time.perf_counter()
         payload = await request.json()
         fut: asyncio.Future = asyncio.get_event_loop().create_future()
