@@ -1,0 +1,78 @@
+"""Property-based tests (hypothesis) for the sampling plan and WLS solver —
+the algorithmic invariants hold for arbitrary (M, nsamples) and arbitrary
+well-posed solve inputs, not just the benchmark shapes."""
+import numpy as np
+import pytest
+from hypothesis import given, settings, strategies as st
+
+from distributedkernelshap_amd.core.sampler import plan_coalitions, sample_masks
+from distributedkernelshap_amd.core.solver import solve_wls
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    m=st.integers(min_value=2, max_value=40),
+    budget=st.integers(min_value=8, max_value=5000),
+    idx=st.integers(min_value=0, max_value=10_000),
+)
+def test_sampler_invariants(m, budget, idx):
+    plan = plan_coalitions(m, nsamples=budget)
+    masks, w = sample_masks(plan, seed=0, instance_index=idx)
+    # shape and budget respected (incl. the 2^m-2 cap)
+    cap = 2 ** m - 2 if m <= 30 else budget
+    assert masks.shape == (min(budget, cap), m)
+    # no empty or full coalitions, ever
+    sizes = masks.sum(axis=1)
+    assert sizes.min() >= 1 and sizes.max() <= m - 1
+    # kernel weights: positive, normalised
+    assert (w > 0).all()
+    assert np.isclose(w.sum(), 1.0)
+    # enumerated block contains no duplicates
+    ne = plan.enum_masks.shape[0]
+    if ne:
+        uniq = np.unique(masks[:ne], axis=0)
+        assert uniq.shape[0] == ne
+    # determinism
+    masks2, w2 = sample_masks(plan, seed=0, instance_index=idx)
+    assert np.array_equal(masks, masks2) and np.array_equal(w, w2)
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    m=st.integers(min_value=2, max_value=24),
+    n_out=st.integers(min_value=1, max_value=3),
+    seed=st.integers(min_value=0, max_value=1000),
+)
+def test_solver_constraint_and_additivity(m, n_out, seed):
+    rng = np.random.Generator(np.random.Philox(key=[seed, 1]))
+    s = max(4 * m, 64)
+    masks = (rng.random((s, m)) > 0.5).astype(np.uint8)
+    masks[masks.sum(axis=1) == 0, 0] = 1
+    masks[masks.sum(axis=1) == m, -1] = 0
+    kw = rng.random(s) + 0.05
+    ey = rng.normal(size=(s, n_out))
+    total = rng.normal(size=(n_out,))
+    phi = solve_wls(masks, kw, ey, total)
+    # local accuracy holds by construction for every output
+    assert np.allclose(phi.sum(axis=0), total, atol=1e-8)
+    # linearity: solving a*ey + b*ey2 == a*phi(ey) + b*phi(ey2) with totals
+    ey2 = rng.normal(size=(s, n_out))
+    total2 = rng.normal(size=(n_out,))
+    a, b = 0.7, -1.3
+    phi2 = solve_wls(masks, kw, ey2, total2)
+    phi_mix = solve_wls(masks, kw, a * ey + b * ey2, a * total + b * total2)
+    assert np.allclose(phi_mix, a * phi + b * phi2, atol=1e-6)
+
+
+@settings(max_examples=20, deadline=None)
+@given(m=st.integers(min_value=2, max_value=11))
+def test_full_enumeration_recovers_additive_game(m):
+    """For any fully-enumerable M (2^m-2 <= 2m+2048 up to m=11), the solve
+    is exact on additive games."""
+    plan = plan_coalitions(m)  # capped at 2^m - 2 -> full enumeration
+    assert plan.n_random == 0
+    masks, kw = sample_masks(plan, 0, 0)
+    contrib = np.linspace(-2, 3, m)
+    ey = (masks @ contrib).reshape(-1, 1)
+    phi = solve_wls(masks, kw, ey, np.array([contrib.sum()]))
+    assert np.allclose(phi[:, 0], contrib, atol=1e-6)
