@@ -223,3 +223,17 @@ def test_explain_checkpointed_resume(tmp_path, problem):
 
     with open(out + "/manifest.json") as f:
         assert len(_json.load(f)["done"]) == 3
+
+
+def test_shard_bounds_partition():
+    from distributedkernelshap_amd.parallel import shard_bounds
+
+    for n in (0, 1, 5, 7, 8, 2560, 1000003):
+        for world in (1, 2, 3, 7, 8):
+            spans = [shard_bounds(n, r, world) for r in range(world)]
+            # contiguous exact partition, balanced within 1
+            assert spans[0][0] == 0 and spans[-1][1] == n
+            for (a, b), (c, d) in zip(spans, spans[1:]):
+                assert b == c
+            sizes = [b - a for a, b in spans]
+            assert max(sizes) - min(sizes) <= 1
