@@ -104,6 +104,10 @@ def main() -> None:
     p.add_argument("--background", type=int, default=None)
     p.add_argument("--nsamples", type=int, default=None)
     p.add_argument("--device", default="auto", choices=["auto", "cuda", "cpu"])
+    p.add_argument("--dtype", default="fp32", choices=["fp32", "bf16x2", "bf16"],
+                   help="GPU predict compute mode: fp32 MFMA (default), "
+                        "bf16 matrix cores with hi+lo split (fp32-grade), "
+                        "or plain bf16 (fastest)")
     args = p.parse_args()
     defaults = CONFIG_DEFAULTS[args.config]
     if args.instances is None:
@@ -142,8 +146,11 @@ def main() -> None:
         pred = LinearPredictor(W, b_, pred.activation)
     background = broadcast_array(background)
 
+    from distributedkernelshap_amd.config import KernelConfig
+
     engine = KernelShapEngine(
-        pred, background, groups=groups, link="logit", seed=0, device=device
+        pred, background, groups=groups, link="logit", seed=0, device=device,
+        kernels=KernelConfig(predict_dtype=args.dtype),
     )
 
     if use_cuda:
@@ -221,7 +228,9 @@ def main() -> None:
             "scaling": "weak",
             "vs_baseline": (value / BASELINE_EXPL_PER_S
                             if args.config == "adult" else None),
-            "dtype": "fp32",
+            "dtype": ("fp32" if args.dtype == "fp32" or device == "cpu"
+                      else ("bf16x2 (hi+lo split, fp32-grade)"
+                            if args.dtype == "bf16x2" else "bf16")),
             "data": "synthetic",
             "config": {
                 "model": model_desc,

@@ -69,6 +69,10 @@ class KernelConfig:
 
     wls_mode: str = "auto"  # auto | mfma | generic | torch
     fused_predict: bool = True
+    # fp32: f32 MFMA (default).  bf16x2: bf16 matrix cores with a hi+lo
+    # split B operand (error ~2^-16, fp32-grade, ~6x less MFMA issue time).
+    # bf16: single bf16 image (fastest, ~0.4% relative ey error).
+    predict_dtype: str = "fp32"  # fp32 | bf16x2 | bf16
     synth_chunk_rows: int = 1 << 19
 
 

@@ -270,6 +270,34 @@ class GpuKernelShap:
         )
         return ey
 
+    def _ey_fused_bf16(self, masks, X_dev, varying, vidx_t=None, packed=None):
+        """bf16 matrix-core predict (predict_dtype 'bf16x2'/'bf16'): one
+        v_mfma_f32_16x16x32_bf16 spans the whole K dim; 'bf16x2' adds a lo
+        correction image (B = hi + lo) for fp32-grade accuracy."""
+        t = self.torch
+        b, s, m = masks.shape
+        npad = (self.N + 15) // 16 * 16
+        split = 2 if self.engine.kernels.predict_dtype == "bf16x2" else 1
+        masksB = self._buf("masksB", (b, s, 32), t.bfloat16)
+        self.ext.expand_masks_bf16(masks, masksB, packed)
+        diff = self._diff_tensor(X_dev, varying, vidx_t)     # (b, o, m, N) f32
+        diffB = self._buf("diffB", (b, split, self.n_out, npad, 40), t.bfloat16)
+        diffB.zero_()
+        dkn = diff.permute(0, 1, 3, 2)                        # (b, o, N, m)
+        hi = dkn.bfloat16()
+        diffB[:, 0, :, : self.N, :m] = hi
+        if split == 2:
+            diffB[:, 1, :, : self.N, :m] = (dkn - hi.float()).bfloat16()
+        base = t.zeros(self.n_out, npad, device=self.device)
+        base[:, : self.N] = self.baseN.T
+        wbg = t.zeros(npad, device=self.device)
+        wbg[: self.N] = self.bg_w
+        ey = self._buf("ey", (b, s, self.n_out))
+        self.ext.fused_predict_bf16(
+            masksB, diffB, base, wbg, ey, self.linear["act"], m
+        )
+        return ey
+
     def _ey_linear_torch(self, masks, X_dev, varying, s_chunk=4096):
         """Library-GEMM fallback for shapes beyond the fused kernel's limits
         (stress configs: M>64 or N>128). ey = reduce(act(mask @ diff + base))."""
@@ -365,11 +393,16 @@ class GpuKernelShap:
         total = (lfx - lfnull64[None, :]).float().contiguous()
         lfnull = lfnull64.float()
         masks, kw = self._device_masks(plan, np.arange(b), ids_dev=ids_dev)
-        mpad = max(4, (m + 3) // 4 * 4)
+        kc = self.engine.kernels
         packed = self._buf("packed", (b, plan.nsamples), t.int64)
-        masksT = self._buf("masksT", (b, mpad, plan.nsamples))
-        self.ext.transpose_masks(masks, masksT, packed)
-        ey = self._ey_fused_linear(masks, masksT, X_dev, varying, vidx_t)
+        if (kc.predict_dtype in ("bf16", "bf16x2") and m <= 32
+                and self.n_out in (1, 2, 4)):
+            ey = self._ey_fused_bf16(masks, X_dev, varying, vidx_t, packed=packed)
+        else:
+            mpad = max(4, (m + 3) // 4 * 4)
+            masksT = self._buf("masksT", (b, mpad, plan.nsamples))
+            self.ext.transpose_masks(masks, masksT, packed)
+            ey = self._ey_fused_linear(masks, masksT, X_dev, varying, vidx_t)
         if self.link_name == "identity":
             ey_adj = ey.sub_(lfnull[None, None, :])
         else:
@@ -550,7 +583,13 @@ class GpuKernelShap:
             if self.linear is not None:
                 mpad = max(4, (m + 3) // 4 * 4)
                 npad = (self.N + 15) // 16 * 16
-                if (kc.fused_predict and mpad <= 64 and npad <= 128
+                use_bf16 = (
+                    kc.predict_dtype in ("bf16", "bf16x2")
+                    and m <= 32 and npad <= 128 and self.n_out in (1, 2, 4)
+                )
+                if kc.fused_predict and use_bf16:
+                    ey = self._ey_fused_bf16(masks, sub_X, varying, packed=packed)
+                elif (kc.fused_predict and mpad <= 64 and npad <= 128
                         and self.n_out in (1, 2, 4)):
                     masksT = self._buf("masksT", (len(ids), mpad, plan.nsamples))
                     self.ext.transpose_masks(masks, masksT, packed)

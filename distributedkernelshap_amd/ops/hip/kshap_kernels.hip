@@ -420,6 +420,228 @@ extern "C" int launch_fused_predict_linear(
 }
 
 // ------------------------------------------------------------------------- //
+// K3-K6 fused, bf16 matrix cores: one v_mfma_f32_16x16x32_bf16 covers the
+// whole K dimension (Mpad <= 32) per column tile.  The A operand (coalition
+// masks) is EXACT in bf16 (0/1); the B operand is either a single bf16
+// image (predict_dtype='bf16') or a hi+lo split pair (predict_dtype=
+// 'bf16x2': B = hi + (B - hi), two MFMAs, error ~2^-16 — fp32-grade).
+// LDS B layout is k-major [o][n][KSTRIDE] so each lane's 8-element k-block
+// is one ds_read_b128; KSTRIDE=40 staggers the 16-lane groups across banks.
+// ------------------------------------------------------------------------- //
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+#define KSTRIDE_BF 40   // 32 k slots + 8 pad: (20*n)%64 distinct for n=0..15
+
+__global__ void expand_masks_bf16_kernel(
+    const uint8_t* __restrict__ masks,  // (B, S, M)
+    __bf16* __restrict__ masksB,        // (B, S, 32) zero-padded k >= M
+    uint64_t* __restrict__ packed,      // (B, S) or null
+    int B, int S, int M)
+{
+    const int b = blockIdx.y;
+    const int sq = blockIdx.x * blockDim.x + threadIdx.x;
+    if (sq >= S) return;
+    const uint8_t* row = masks + ((size_t)b * S + sq) * M;
+    __bf16* out = masksB + ((size_t)b * S + sq) * 32;
+    uint64_t bits = 0ull;
+    for (int k = 0; k < 32; ++k) {
+        uint8_t v = (k < M) ? (row[k] & 1) : 0;
+        out[k] = (__bf16)(float)v;
+        bits |= ((uint64_t)v) << k;
+    }
+    if (packed) packed[(size_t)b * S + sq] = bits;
+}
+
+extern "C" void launch_expand_masks_bf16(
+    const uint8_t* masks, uint16_t* masksB, uint64_t* packed, int B, int S,
+    int M, hipStream_t stream)
+{
+    dim3 grid((S + 255) / 256, B), block(256);
+    expand_masks_bf16_kernel<<<grid, block, 0, stream>>>(
+        masks, reinterpret_cast<__bf16*>(masksB), packed, B, S, M);
+}
+
+template <int NOUT, int ACT, int NT, int SPLIT>  // SPLIT: 1 = hi only, 2 = hi+lo
+__global__ __launch_bounds__(256)
+void fused_predict_bf16_kernel(
+    const __bf16* __restrict__ masksB,  // (B, S, 32)
+    const __bf16* __restrict__ diffB,   // (B, SPLIT, NOUT, NT*16, KSTRIDE_BF)
+    const float* __restrict__ base,     // (NOUT, NT*16)
+    const float* __restrict__ wbg,      // (NT*16)
+    float* __restrict__ ey,             // (B, S, NOUT)
+    int B, int S)
+{
+    constexpr int NPAD = NT * 16;
+    const int n_stiles = (S + S_TILE - 1) / S_TILE;
+    const int b = blockIdx.x / n_stiles;
+    const int stile = blockIdx.x % n_stiles;
+    const int s0 = stile * S_TILE;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wave = tid >> 6;
+
+    extern __shared__ float lds[];
+    __bf16* diff_lds = (__bf16*)lds;                 // SPLIT*NOUT*NPAD*KSTRIDE_BF
+    float* base_lds = lds + (SPLIT * NOUT * NPAD * KSTRIDE_BF + 1) / 2;
+    float* wbg_lds = base_lds + NOUT * NPAD;
+
+    const __bf16* dsrc = diffB + (size_t)b * SPLIT * NOUT * NPAD * KSTRIDE_BF;
+    for (int idx = tid; idx < SPLIT * NOUT * NPAD * KSTRIDE_BF / 8; idx += 256)
+        ((bf16x8*)diff_lds)[idx] = ((const bf16x8*)dsrc)[idx];
+    for (int idx = tid; idx < NOUT * NPAD; idx += 256) base_lds[idx] = base[idx];
+    for (int idx = tid; idx < NPAD; idx += 256) wbg_lds[idx] = wbg[idx];
+    __syncthreads();
+
+    const int swave = wave * 16;
+    const int arow = lane & 15;          // A row (s) / B col (n)
+    const int akb = lane >> 4;           // k-block 0..3 (8 elements each)
+    const __bf16* mlane = masksB + ((size_t)b * S + swave + arow) * 32 + akb * 8;
+    // per-lane LDS base for B fragments: [split][o][n=ct*16+arow][k=akb*8]
+    const __bf16* dlane = diff_lds + (size_t)arow * KSTRIDE_BF + akb * 8;
+
+    for (int sub = 0; sub < S_TILE / S_SUB; ++sub) {
+        const int ssub0 = s0 + sub * S_SUB;
+        if (ssub0 >= S) break;
+        const int srow = ssub0 + swave + arow;
+        const bool svalid = srow < S;
+        bf16x8 a;
+        if (svalid)
+            a = *(const bf16x8*)(mlane + (size_t)ssub0 * 32);
+        else
+#pragma unroll
+            for (int j = 0; j < 8; ++j) a[j] = (__bf16)0.0f;
+
+        float partial[NOUT][4];
+#pragma unroll
+        for (int o = 0; o < NOUT; ++o)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) partial[o][r] = 0.0f;
+
+#pragma unroll
+        for (int ct = 0; ct < NT; ++ct) {
+            f32x4 acc[NOUT];
+#pragma unroll
+            for (int o = 0; o < NOUT; ++o) {
+                acc[o] = (f32x4){0, 0, 0, 0};
+#pragma unroll
+                for (int sp = 0; sp < SPLIT; ++sp) {
+                    bf16x8 bv = *(const bf16x8*)(
+                        dlane + ((size_t)(sp * NOUT + o) * NPAD + ct * 16) * KSTRIDE_BF);
+                    acc[o] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a, bv, acc[o], 0, 0, 0);
+                }
+            }
+            const int n = ct * 16 + arow;
+            float wn = wbg_lds[n];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float z[NOUT];
+#pragma unroll
+                for (int o = 0; o < NOUT; ++o) z[o] = acc[o][r] + base_lds[o * NPAD + n];
+                if (ACT == 1) {
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) z[o] = 1.0f / (1.0f + __expf(-z[o]));
+                } else if (ACT == 2 && NOUT == 2) {
+                    float p1 = 1.0f / (1.0f + __expf(z[0] - z[1]));
+                    z[0] = 1.0f - p1;
+                    z[1] = p1;
+                } else if (ACT == 2) {
+                    float mx = z[0];
+#pragma unroll
+                    for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
+                    float sum = 0.0f;
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) { z[o] = __expf(z[o] - mx); sum += z[o]; }
+                    float inv = 1.0f / sum;
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) z[o] *= inv;
+                }
+#pragma unroll
+                for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * z[o];
+            }
+        }
+#pragma unroll
+        for (int o = 0; o < NOUT; ++o)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float v = partial[o][r];
+                v += __shfl_xor(v, 1);
+                v += __shfl_xor(v, 2);
+                v += __shfl_xor(v, 4);
+                v += __shfl_xor(v, 8);
+                partial[o][r] = v;
+            }
+        if (arow == 0) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int ss = ssub0 + swave + akb * 4 + r;
+                if (ss < S) {
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o)
+                        ey[((size_t)b * S + ss) * NOUT + o] = partial[o][r];
+                }
+            }
+        }
+    }
+}
+
+template <int NOUT, int ACT>
+static void launch_fused_bf16_nt(
+    const __bf16* masksB, const __bf16* diffB, const float* base,
+    const float* wbg, float* ey, int B, int S, int Npad, int split,
+    hipStream_t stream)
+{
+    const int n_stiles = (S + S_TILE - 1) / S_TILE;
+    dim3 grid(B * n_stiles), block(256);
+    size_t lds = ((size_t)split * NOUT * Npad * KSTRIDE_BF * 2 + 2)
+                 + (size_t)(NOUT * Npad + Npad) * 4;
+    lds = (lds + 3) / 4 * 4 + 4;
+#define KSHAP_BF_CASE(NTV, SPL) \
+    if (Npad / 16 == NTV && split == SPL) { \
+        fused_predict_bf16_kernel<NOUT, ACT, NTV, SPL><<<grid, block, lds, stream>>>( \
+            masksB, diffB, base, wbg, ey, B, S); \
+        return; \
+    }
+    KSHAP_BF_CASE(1,1) KSHAP_BF_CASE(2,1) KSHAP_BF_CASE(3,1) KSHAP_BF_CASE(4,1)
+    KSHAP_BF_CASE(5,1) KSHAP_BF_CASE(6,1) KSHAP_BF_CASE(7,1) KSHAP_BF_CASE(8,1)
+    KSHAP_BF_CASE(1,2) KSHAP_BF_CASE(2,2) KSHAP_BF_CASE(3,2) KSHAP_BF_CASE(4,2)
+    KSHAP_BF_CASE(5,2) KSHAP_BF_CASE(6,2) KSHAP_BF_CASE(7,2) KSHAP_BF_CASE(8,2)
+#undef KSHAP_BF_CASE
+}
+
+template <int NOUT>
+static void launch_fused_bf16_act(
+    const __bf16* masksB, const __bf16* diffB, const float* base,
+    const float* wbg, float* ey, int B, int S, int Npad, int act, int split,
+    hipStream_t stream)
+{
+    switch (act) {
+        case 0: launch_fused_bf16_nt<NOUT, 0>(masksB, diffB, base, wbg, ey, B, S, Npad, split, stream); break;
+        case 1: launch_fused_bf16_nt<NOUT, 1>(masksB, diffB, base, wbg, ey, B, S, Npad, split, stream); break;
+        default: launch_fused_bf16_nt<NOUT, 2>(masksB, diffB, base, wbg, ey, B, S, Npad, split, stream); break;
+    }
+}
+
+extern "C" int launch_fused_predict_bf16(
+    const uint16_t* masksB_u, const uint16_t* diffB_u, const float* base,
+    const float* wbg, float* ey, int B, int S, int M, int Npad, int n_out,
+    int act, int split, hipStream_t stream)
+{
+    if (M > 32 || Npad % 16 != 0 || Npad / 16 > 8 || split < 1 || split > 2)
+        return -1;
+    const __bf16* masksB = reinterpret_cast<const __bf16*>(masksB_u);
+    const __bf16* diffB = reinterpret_cast<const __bf16*>(diffB_u);
+    switch (n_out) {
+        case 1: launch_fused_bf16_act<1>(masksB, diffB, base, wbg, ey, B, S, Npad, act, split, stream); break;
+        case 2: launch_fused_bf16_act<2>(masksB, diffB, base, wbg, ey, B, S, Npad, act, split, stream); break;
+        case 4: launch_fused_bf16_act<4>(masksB, diffB, base, wbg, ey, B, S, Npad, act, split, stream); break;
+        default: return -1;
+    }
+    return 0;
+}
+
+// ------------------------------------------------------------------------- //
 // K3': explicit masked-background synthesis for the torch-predictor path.
 // out[(s - s_lo)*N + n, d] = mask[b, s, group(d)] ? x[b, d] : bg[n, d]
 // Coalesced over d; one workgroup covers one (s, n) row-pair block.

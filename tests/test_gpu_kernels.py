@@ -552,3 +552,47 @@ def test_gpu_pipeline_bitwise_deterministic():
     a, b = run(), run()
     for o in range(2):
         assert np.array_equal(a[o], b[o])
+
+
+# ----------------------------------------------------------------------- #
+# bf16 matrix-core predict (opt-in predict_dtype)
+
+def _bf16_engine(dtype):
+    from distributedkernelshap_amd.config import KernelConfig
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor, make_adult_like
+
+    data = make_adult_like(n_instances=32, n_background=100, seed=9)
+    pred = LinearPredictor.random(data.X.shape[1], 2, seed=9)
+    eng = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cuda", kernels=KernelConfig(predict_dtype=dtype),
+    )
+    return eng, data, pred
+
+
+def test_bf16x2_split_matches_fp32():
+    """hi+lo split bf16 path is fp32-grade (error ~2^-16 on the B operand)."""
+    eng32, data, pred = _bf16_engine("fp32")
+    engb2, _, _ = _bf16_engine("bf16x2")
+    sv32 = eng32.shap_values(data.X)
+    svb2 = engb2.shap_values(data.X)
+    for o in range(2):
+        err = np.abs(sv32[o] - svb2[o]).max()
+        assert err < 2e-3, err
+
+
+def test_bf16_single_local_accuracy():
+    """Plain bf16 path: coarser ey (~0.4% rel) but local accuracy still holds
+    exactly (constrained solve) and phi stays close to fp32."""
+    from distributedkernelshap_amd.core.links import logit
+
+    engb, data, pred = _bf16_engine("bf16")
+    sv = engb.shap_values(data.X)
+    fx = logit(pred(data.X))
+    for o in range(2):
+        total = sv[o].sum(axis=1) + engb.expected_value[o]
+        assert np.abs(total - fx[:, o]).max() < 1e-3
+    eng32, _, _ = _bf16_engine("fp32")
+    sv32 = eng32.shap_values(data.X)
+    assert np.abs(sv[0] - sv32[0]).max() < 0.1
