@@ -437,29 +437,30 @@ __global__ void expand_masks_bf16_kernel(
     const uint8_t* __restrict__ masks,  // (B, S, M)
     __bf16* __restrict__ masksB,        // (B, S, 32) zero-padded k >= M
     uint64_t* __restrict__ packed,      // (B, S) or null
-    int B, int S, int M)
+    size_t n_rows, int M)
 {
-    const int b = blockIdx.y;
-    const int sq = blockIdx.x * blockDim.x + threadIdx.x;
-    if (sq >= S) return;
-    const uint8_t* row = masks + ((size_t)b * S + sq) * M;
-    __bf16* out = masksB + ((size_t)b * S + sq) * 32;
-    uint64_t bits = 0ull;
-    for (int k = 0; k < 32; ++k) {
-        uint8_t v = (k < M) ? (row[k] & 1) : 0;
-        out[k] = (__bf16)(float)v;
-        bits |= ((uint64_t)v) << k;
-    }
-    if (packed) packed[(size_t)b * S + sq] = bits;
+    // one thread per OUTPUT element (coalesced read + write); a 64-lane
+    // wave covers two mask rows, the packed bits fall out of a ballot
+    const size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n_rows * 32) return;
+    const size_t row = idx >> 5;
+    const int k = idx & 31;
+    const int lane = threadIdx.x & (WAVE - 1);
+    uint8_t v = (k < M) ? (masks[row * M + k] & 1) : 0;
+    masksB[idx] = (__bf16)(float)v;
+    uint64_t bal = __ballot(v != 0);
+    if (packed && k == 0)
+        packed[row] = (lane < 32) ? (bal & 0xffffffffull) : (bal >> 32);
 }
 
 extern "C" void launch_expand_masks_bf16(
     const uint8_t* masks, uint16_t* masksB, uint64_t* packed, int B, int S,
     int M, hipStream_t stream)
 {
-    dim3 grid((S + 255) / 256, B), block(256);
+    size_t n = (size_t)B * S * 32;
+    dim3 grid((unsigned)((n + 255) / 256)), block(256);
     expand_masks_bf16_kernel<<<grid, block, 0, stream>>>(
-        masks, reinterpret_cast<__bf16*>(masksB), packed, B, S, M);
+        masks, reinterpret_cast<__bf16*>(masksB), packed, (size_t)B * S, M);
 }
 
 template <int NOUT, int ACT, int NT, int SPLIT>  // SPLIT: 1 = hi only, 2 = hi+lo
