@@ -131,13 +131,15 @@ class GpuKernelShap:
     def enable_tracing(self, on: bool = True) -> None:
         self.trace = {} if on else None
 
-    def _buf(self, name, shape, dtype=None):
+    def _buf(self, name, shape, dtype=None, zeroed=False):
         t = self.torch
         dtype = dtype or t.float32
         key = (name, tuple(shape), dtype)   # per-shape: captured hipGraphs
         buf = self._ws.get(key)             # hold raw pointers to these
         if buf is None:
-            buf = t.empty(*shape, dtype=dtype, device=self.device)
+            buf = (t.zeros if zeroed else t.empty)(
+                *shape, dtype=dtype, device=self.device
+            )
             self._ws[key] = buf
         return buf
 
@@ -257,9 +259,13 @@ class GpuKernelShap:
         b, s, m = masks.shape
         mpad = masksT.shape[1]
         npad = (self.N + 15) // 16 * 16
-        diff = self._buf("diff", (b, self.n_out, mpad, npad))
-        diff.zero_()
-        diff[:, :, :m, : self.N] = self._diff_tensor(X_dev, varying, vidx_t)
+        if vidx_t is None:
+            vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
+        # pad slots are zeroed once at allocation and never written after
+        # (buffer name keyed by m so a different varying count cannot see
+        # stale values)
+        diff = self._buf(f"diff{m}", (b, self.n_out, mpad, npad), zeroed=True)
+        self.ext.build_diff_f32(self._x_part(X_dev), self.bg_part, vidx_t, diff)
         base = t.zeros(self.n_out, npad, device=self.device)
         base[:, : self.N] = self.baseN.T
         wbg = t.zeros(npad, device=self.device)
@@ -280,14 +286,13 @@ class GpuKernelShap:
         split = 2 if self.engine.kernels.predict_dtype == "bf16x2" else 1
         masksB = self._buf("masksB", (b, s, 32), t.bfloat16)
         self.ext.expand_masks_bf16(masks, masksB, packed)
-        diff = self._diff_tensor(X_dev, varying, vidx_t)     # (b, o, m, N) f32
-        diffB = self._buf("diffB", (b, split, self.n_out, npad, 40), t.bfloat16)
-        diffB.zero_()
-        dkn = diff.permute(0, 1, 3, 2)                        # (b, o, N, m)
-        hi = dkn.bfloat16()
-        diffB[:, 0, :, : self.N, :m] = hi
-        if split == 2:
-            diffB[:, 1, :, : self.N, :m] = (dkn - hi.float()).bfloat16()
+        if vidx_t is None:
+            vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
+        diffB = self._buf(
+            f"diffB{m}", (b, split, self.n_out, npad, 40), t.bfloat16,
+            zeroed=True,
+        )
+        self.ext.build_diff_bf16(self._x_part(X_dev), self.bg_part, vidx_t, diffB)
         base = t.zeros(self.n_out, npad, device=self.device)
         base[:, : self.N] = self.baseN.T
         wbg = t.zeros(npad, device=self.device)

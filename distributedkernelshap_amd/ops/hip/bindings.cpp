@@ -34,6 +34,14 @@ extern "C" int launch_fused_predict_bf16(
     const float* base, const float* wbg, float* ey, int B, int S, int M,
     int Npad, int n_out, int act, int split, hipStream_t stream);
 
+extern "C" void launch_build_diff_f32(
+    const float* xp, const float* bgp, const int64_t* vidx, float* out,
+    int B, int G, int O, int N, int m, int Mpad, int Npad, hipStream_t stream);
+
+extern "C" void launch_build_diff_bf16(
+    const float* xp, const float* bgp, const int64_t* vidx, uint16_t* out,
+    int B, int G, int O, int N, int m, int Npad, int split, hipStream_t stream);
+
 extern "C" int launch_wls_solve(
     const uint8_t* masks, const uint64_t* packed, const float* kw,
     const float* ey_adj, const float* total, float* phi, int B, int S, int M,
@@ -176,6 +184,32 @@ void fused_predict_bf16(
     TORCH_CHECK(rc == 0, "fused_predict_bf16: unsupported shape (M<=32, Npad<=128, n_out in {1,2,4}, split in {1,2})");
 }
 
+void build_diff_f32(torch::Tensor xp, torch::Tensor bgp, torch::Tensor vidx,
+                    torch::Tensor out) {
+    CHECK_DEV(xp); CHECK_DEV(bgp); CHECK_DEV(vidx); CHECK_DEV(out);
+    int B = xp.size(0), G = xp.size(1), O = xp.size(2);
+    int N = bgp.size(0), m = vidx.size(0);
+    int Mpad = out.size(2), Npad = out.size(3);
+    TORCH_CHECK(out.size(0) == B && out.size(1) == O && m <= Mpad && N <= Npad);
+    launch_build_diff_f32(
+        xp.data_ptr<float>(), bgp.data_ptr<float>(), vidx.data_ptr<int64_t>(),
+        out.data_ptr<float>(), B, G, O, N, m, Mpad, Npad, current_stream());
+}
+
+void build_diff_bf16(torch::Tensor xp, torch::Tensor bgp, torch::Tensor vidx,
+                     torch::Tensor out) {
+    CHECK_DEV(xp); CHECK_DEV(bgp); CHECK_DEV(vidx); CHECK_DEV(out);
+    TORCH_CHECK(out.dtype() == torch::kBFloat16 && out.size(4) == 40);
+    int B = xp.size(0), G = xp.size(1), O = xp.size(2);
+    int N = bgp.size(0), m = vidx.size(0);
+    int split = out.size(1), Npad = out.size(3);
+    TORCH_CHECK(out.size(0) == B && out.size(2) == O && N <= Npad && m <= 32);
+    launch_build_diff_bf16(
+        xp.data_ptr<float>(), bgp.data_ptr<float>(), vidx.data_ptr<int64_t>(),
+        reinterpret_cast<uint16_t*>(out.data_ptr<at::BFloat16>()), B, G, O, N,
+        m, Npad, split, current_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -183,6 +217,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "Philox coalition sampling (K2)");
     m.def("fused_predict_linear", &fused_predict_linear,
           "MFMA fused mask@diff GEMM + activation + background reduce (K3-K6)");
+    m.def("build_diff_f32", &build_diff_f32,
+          "diff image in the f32 fused-kernel layout");
+    m.def("build_diff_bf16", &build_diff_bf16,
+          "hi(+lo) diff images in the bf16 fused-kernel layout");
     m.def("expand_masks_bf16", &expand_masks_bf16,
           "masks u8 -> (B,S,32) bf16 (+ packed u64) for the bf16 MFMA path",
           pybind11::arg("masks"), pybind11::arg("masksB"),

@@ -643,6 +643,75 @@ extern "C" int launch_fused_predict_bf16(
 }
 
 // ------------------------------------------------------------------------- //
+// diff-image builders: diff[k, n] = x_part[b, vidx[k], o] - bg_part[n,
+// vidx[k], o], written directly in the fused kernels\' operand layouts
+// (replaces a chain of strided torch scatter kernels inside the graph).
+// ------------------------------------------------------------------------- //
+
+__global__ void build_diff_f32_kernel(
+    const float* __restrict__ xp,    // (B, G, O)
+    const float* __restrict__ bgp,   // (N, G, O)
+    const int64_t* __restrict__ vidx,  // (m,)
+    float* __restrict__ out,         // (B, O, Mpad, Npad) — pad slots stay 0
+    int G, int O, int N, int m, int Mpad, int Npad, size_t total)
+{
+    size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= total) return;       // total = B*O*m*N, n fastest
+    const int n = idx % N;
+    const int k = (idx / N) % m;
+    const int o = (idx / ((size_t)N * m)) % O;
+    const size_t b = idx / ((size_t)N * m * O);
+    const int64_t g = vidx[k];
+    float v = xp[((size_t)b * G + g) * O + o] - bgp[((size_t)n * G + g) * O + o];
+    out[(((size_t)b * O + o) * Mpad + k) * Npad + n] = v;
+}
+
+extern "C" void launch_build_diff_f32(
+    const float* xp, const float* bgp, const int64_t* vidx, float* out,
+    int B, int G, int O, int N, int m, int Mpad, int Npad, hipStream_t stream)
+{
+    size_t total = (size_t)B * O * m * N;
+    build_diff_f32_kernel<<<dim3((unsigned)((total + 255) / 256)), dim3(256), 0,
+                            stream>>>(xp, bgp, vidx, out, G, O, N, m, Mpad,
+                                      Npad, total);
+}
+
+__global__ void build_diff_bf16_kernel(
+    const float* __restrict__ xp,    // (B, G, O)
+    const float* __restrict__ bgp,   // (N, G, O)
+    const int64_t* __restrict__ vidx,  // (m,)
+    __bf16* __restrict__ out,        // (B, SPLIT, O, Npad, KSTRIDE_BF)
+    int G, int O, int N, int m, int Npad, int split, size_t total)
+{
+    size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= total) return;       // total = B*O*N*m, k fastest
+    const int k = idx % m;
+    const int n = (idx / m) % N;
+    const int o = (idx / ((size_t)m * N)) % O;
+    const size_t b = idx / ((size_t)m * N * O);
+    const int64_t g = vidx[k];
+    float v = xp[((size_t)b * G + g) * O + o] - bgp[((size_t)n * G + g) * O + o];
+    __bf16 hi = (__bf16)v;
+    size_t base = ((((size_t)b * split) * O + o) * Npad + n) * KSTRIDE_BF + k;
+    out[base] = hi;
+    if (split == 2) {
+        size_t lobase = ((((size_t)b * split + 1) * O + o) * Npad + n) * KSTRIDE_BF + k;
+        out[lobase] = (__bf16)(v - (float)hi);
+    }
+}
+
+extern "C" void launch_build_diff_bf16(
+    const float* xp, const float* bgp, const int64_t* vidx, uint16_t* out,
+    int B, int G, int O, int N, int m, int Npad, int split, hipStream_t stream)
+{
+    size_t total = (size_t)B * O * N * m;
+    build_diff_bf16_kernel<<<dim3((unsigned)((total + 255) / 256)), dim3(256),
+                             0, stream>>>(
+        xp, bgp, vidx, reinterpret_cast<__bf16*>(out), G, O, N, m, Npad,
+        split, total);
+}
+
+// ------------------------------------------------------------------------- //
 // K3': explicit masked-background synthesis for the torch-predictor path.
 // out[(s - s_lo)*N + n, d] = mask[b, s, group(d)] ? x[b, d] : bg[n, d]
 // Coalesced over d; one workgroup covers one (s, n) row-pair block.
