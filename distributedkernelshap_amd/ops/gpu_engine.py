@@ -284,8 +284,8 @@ class GpuKernelShap:
         b, s, m = masks.shape
         npad = (self.N + 15) // 16 * 16
         split = 2 if self.engine.kernels.predict_dtype == "bf16x2" else 1
-        masksB = self._buf("masksB", (b, s, 32), t.bfloat16)
-        self.ext.expand_masks_bf16(masks, masksB, packed)
+        if packed is not None:
+            self.ext.pack_masks(masks, packed)
         if vidx_t is None:
             vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
         diffB = self._buf(
@@ -299,7 +299,7 @@ class GpuKernelShap:
         wbg[: self.N] = self.bg_w
         ey = self._buf("ey", (b, s, self.n_out))
         self.ext.fused_predict_bf16(
-            masksB, diffB, base, wbg, ey, self.linear["act"], m
+            masks, diffB, base, wbg, ey, self.linear["act"]
         )
         return ey
 

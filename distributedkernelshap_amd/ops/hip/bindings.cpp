@@ -25,12 +25,12 @@ extern "C" void launch_synth_chunk(
     int s_hi, hipStream_t stream);
 
 // bf16 device pointers are opaque 16-bit words host-side (extern "C")
-extern "C" void launch_expand_masks_bf16(
-    const uint8_t* masks, uint16_t* masksB, uint64_t* packed, int B,
-    int S, int M, hipStream_t stream);
+extern "C" void launch_pack_masks(
+    const uint8_t* masks, uint64_t* packed, int B, int S, int M,
+    hipStream_t stream);
 
 extern "C" int launch_fused_predict_bf16(
-    const uint16_t* masksB, const uint16_t* diffB,
+    const uint8_t* masksU, const uint16_t* diffB,
     const float* base, const float* wbg, float* ey, int B, int S, int M,
     int Npad, int n_out, int act, int split, hipStream_t stream);
 
@@ -146,41 +146,32 @@ void wls_solve(
     TORCH_CHECK(rc == 0, "wls_solve: unsupported shape (2<=M<=64, n_out<=8)");
 }
 
-void expand_masks_bf16(torch::Tensor masks, torch::Tensor masksB,
-                       c10::optional<torch::Tensor> packed) {
-    CHECK_DEV(masks); CHECK_DEV(masksB);
+void pack_masks(torch::Tensor masks, torch::Tensor packed) {
+    CHECK_DEV(masks); CHECK_DEV(packed);
     TORCH_CHECK(masks.dtype() == torch::kUInt8, "masks must be u8");
-    TORCH_CHECK(masksB.dtype() == torch::kBFloat16, "masksB must be bf16");
     int B = masks.size(0), S = masks.size(1), M = masks.size(2);
-    TORCH_CHECK(M <= 32, "bf16 path supports M <= 32");
-    TORCH_CHECK(masksB.size(0) == B && masksB.size(1) == S && masksB.size(2) == 32,
-                "masksB shape (B,S,32)");
-    uint64_t* pk = nullptr;
-    if (packed.has_value()) {
-        CHECK_DEV(packed.value());
-        TORCH_CHECK(packed->size(0) == B && packed->size(1) == S, "packed shape");
-        pk = reinterpret_cast<uint64_t*>(packed->data_ptr<int64_t>());
-    }
-    launch_expand_masks_bf16(
+    TORCH_CHECK(M <= 32, "pack_masks supports M <= 32");
+    TORCH_CHECK(packed.size(0) == B && packed.size(1) == S, "packed shape");
+    launch_pack_masks(
         masks.data_ptr<uint8_t>(),
-        reinterpret_cast<uint16_t*>(masksB.data_ptr<at::BFloat16>()),
-        pk, B, S, M, current_stream());
+        reinterpret_cast<uint64_t*>(packed.data_ptr<int64_t>()),
+        B, S, M, current_stream());
 }
 
 void fused_predict_bf16(
-    torch::Tensor masksB, torch::Tensor diffB, torch::Tensor base,
-    torch::Tensor wbg, torch::Tensor ey, int64_t act, int64_t m) {
-    CHECK_DEV(masksB); CHECK_DEV(diffB); CHECK_DEV(base); CHECK_DEV(wbg); CHECK_DEV(ey);
-    TORCH_CHECK(masksB.dtype() == torch::kBFloat16 && diffB.dtype() == torch::kBFloat16);
-    int B = masksB.size(0), S = masksB.size(1);
+    torch::Tensor masks, torch::Tensor diffB, torch::Tensor base,
+    torch::Tensor wbg, torch::Tensor ey, int64_t act) {
+    CHECK_DEV(masks); CHECK_DEV(diffB); CHECK_DEV(base); CHECK_DEV(wbg); CHECK_DEV(ey);
+    TORCH_CHECK(masks.dtype() == torch::kUInt8 && diffB.dtype() == torch::kBFloat16);
+    int B = masks.size(0), S = masks.size(1), M = masks.size(2);
     int split = diffB.size(1), n_out = diffB.size(2), Npad = diffB.size(3);
     TORCH_CHECK(diffB.size(0) == B && diffB.size(4) == 40, "diffB (B,split,o,Npad,40)");
     TORCH_CHECK(ey.size(0) == B && ey.size(1) == S && ey.size(2) == n_out, "ey shape");
     int rc = launch_fused_predict_bf16(
-        reinterpret_cast<const uint16_t*>(masksB.data_ptr<at::BFloat16>()),
+        masks.data_ptr<uint8_t>(),
         reinterpret_cast<const uint16_t*>(diffB.data_ptr<at::BFloat16>()),
         base.data_ptr<float>(), wbg.data_ptr<float>(), ey.data_ptr<float>(),
-        B, S, (int)m, Npad, n_out, (int)act, split, current_stream());
+        B, S, M, Npad, n_out, (int)act, split, current_stream());
     TORCH_CHECK(rc == 0, "fused_predict_bf16: unsupported shape (M<=32, Npad<=128, n_out in {1,2,4}, split in {1,2})");
 }
 
@@ -221,12 +212,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "diff image in the f32 fused-kernel layout");
     m.def("build_diff_bf16", &build_diff_bf16,
           "hi(+lo) diff images in the bf16 fused-kernel layout");
-    m.def("expand_masks_bf16", &expand_masks_bf16,
-          "masks u8 -> (B,S,32) bf16 (+ packed u64) for the bf16 MFMA path",
-          pybind11::arg("masks"), pybind11::arg("masksB"),
-          pybind11::arg("packed") = pybind11::none());
+    m.def("pack_masks", &pack_masks,
+          "masks u8 -> packed u64 bits (for the MFMA WLS Gram build)");
     m.def("fused_predict_bf16", &fused_predict_bf16,
-          "bf16 matrix-core fused predict (single or hi+lo split)");
+          "bf16 matrix-core fused predict (single or hi+lo split), A-operand "
+          "converted in-register from the raw u8 masks");
     m.def("transpose_masks", &transpose_masks,
           "masks (B,S,M) u8 -> (B,Mpad,S) f32 (+ packed u64) for coalesced "
           "A staging (K2b)",

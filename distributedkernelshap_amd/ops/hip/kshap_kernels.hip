@@ -433,45 +433,42 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
 #define KSTRIDE_BF 40   // 32 k slots + 8 pad: (20*n)%64 distinct for n=0..15
 
-__global__ void expand_masks_bf16_kernel(
+__global__ void pack_masks_kernel(
     const uint8_t* __restrict__ masks,  // (B, S, M)
-    __bf16* __restrict__ masksB,        // (B, S, 32) zero-padded k >= M
-    uint64_t* __restrict__ packed,      // (B, S) or null
+    uint64_t* __restrict__ packed,      // (B, S)
     size_t n_rows, int M)
 {
-    // one thread per OUTPUT element (coalesced read + write); a 64-lane
-    // wave covers two mask rows, the packed bits fall out of a ballot
+    // thread per (row, k<=31); packed bits fall out of a half-wave ballot
     const size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (idx >= n_rows * 32) return;
     const size_t row = idx >> 5;
     const int k = idx & 31;
     const int lane = threadIdx.x & (WAVE - 1);
     uint8_t v = (k < M) ? (masks[row * M + k] & 1) : 0;
-    masksB[idx] = (__bf16)(float)v;
     uint64_t bal = __ballot(v != 0);
-    if (packed && k == 0)
+    if (k == 0)
         packed[row] = (lane < 32) ? (bal & 0xffffffffull) : (bal >> 32);
 }
 
-extern "C" void launch_expand_masks_bf16(
-    const uint8_t* masks, uint16_t* masksB, uint64_t* packed, int B, int S,
-    int M, hipStream_t stream)
+extern "C" void launch_pack_masks(
+    const uint8_t* masks, uint64_t* packed, int B, int S, int M,
+    hipStream_t stream)
 {
     size_t n = (size_t)B * S * 32;
     dim3 grid((unsigned)((n + 255) / 256)), block(256);
-    expand_masks_bf16_kernel<<<grid, block, 0, stream>>>(
-        masks, reinterpret_cast<__bf16*>(masksB), packed, (size_t)B * S, M);
+    pack_masks_kernel<<<grid, block, 0, stream>>>(
+        masks, packed, (size_t)B * S, M);
 }
 
 template <int NOUT, int ACT, int NT, int SPLIT>  // SPLIT: 1 = hi only, 2 = hi+lo
 __global__ __launch_bounds__(256)
 void fused_predict_bf16_kernel(
-    const __bf16* __restrict__ masksB,  // (B, S, 32)
+    const uint8_t* __restrict__ masksU, // (B, S, M) raw coalition masks
     const __bf16* __restrict__ diffB,   // (B, SPLIT, NOUT, NT*16, KSTRIDE_BF)
     const float* __restrict__ base,     // (NOUT, NT*16)
     const float* __restrict__ wbg,      // (NT*16)
     float* __restrict__ ey,             // (B, S, NOUT)
-    int B, int S)
+    int B, int S, int M)
 {
     constexpr int NPAD = NT * 16;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
@@ -497,7 +494,7 @@ void fused_predict_bf16_kernel(
     const int swave = wave * 16;
     const int arow = lane & 15;          // A row (s) / B col (n)
     const int akb = lane >> 4;           // k-block 0..3 (8 elements each)
-    const __bf16* mlane = masksB + ((size_t)b * S + swave + arow) * 32 + akb * 8;
+    const uint8_t* mlane = masksU + ((size_t)b * S + swave + arow) * M;
     // per-lane LDS base for B fragments: [split][o][n=ct*16+arow][k=akb*8]
     const __bf16* dlane = diff_lds + (size_t)arow * KSTRIDE_BF + akb * 8;
 
@@ -506,12 +503,17 @@ void fused_predict_bf16_kernel(
         if (ssub0 >= S) break;
         const int srow = ssub0 + swave + arow;
         const bool svalid = srow < S;
+        // A fragment converted in-register from the raw u8 masks (exact in
+        // bf16); a dedicated bf16 mask image would cost 1.4 GB/step of
+        // HBM traffic at B=10k
+        const uint8_t* mrow = mlane + (size_t)ssub0 * M;
         bf16x8 a;
-        if (svalid)
-            a = *(const bf16x8*)(mlane + (size_t)ssub0 * 32);
-        else
 #pragma unroll
-            for (int j = 0; j < 8; ++j) a[j] = (__bf16)0.0f;
+        for (int j = 0; j < 8; ++j) {
+            const int k = akb * 8 + j;
+            a[j] = (__bf16)(float)(
+                (svalid && k < M) ? (mrow[k] & 1) : 0);
+        }
 
         float partial[NOUT][4];
 #pragma unroll
@@ -589,8 +591,8 @@ void fused_predict_bf16_kernel(
 
 template <int NOUT, int ACT>
 static void launch_fused_bf16_nt(
-    const __bf16* masksB, const __bf16* diffB, const float* base,
-    const float* wbg, float* ey, int B, int S, int Npad, int split,
+    const uint8_t* masksU, const __bf16* diffB, const float* base,
+    const float* wbg, float* ey, int B, int S, int M, int Npad, int split,
     hipStream_t stream)
 {
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
@@ -601,7 +603,7 @@ static void launch_fused_bf16_nt(
 #define KSHAP_BF_CASE(NTV, SPL) \
     if (Npad / 16 == NTV && split == SPL) { \
         fused_predict_bf16_kernel<NOUT, ACT, NTV, SPL><<<grid, block, lds, stream>>>( \
-            masksB, diffB, base, wbg, ey, B, S); \
+            masksU, diffB, base, wbg, ey, B, S, M); \
         return; \
     }
     KSHAP_BF_CASE(1,1) KSHAP_BF_CASE(2,1) KSHAP_BF_CASE(3,1) KSHAP_BF_CASE(4,1)
@@ -613,30 +615,29 @@ static void launch_fused_bf16_nt(
 
 template <int NOUT>
 static void launch_fused_bf16_act(
-    const __bf16* masksB, const __bf16* diffB, const float* base,
-    const float* wbg, float* ey, int B, int S, int Npad, int act, int split,
-    hipStream_t stream)
+    const uint8_t* masksU, const __bf16* diffB, const float* base,
+    const float* wbg, float* ey, int B, int S, int M, int Npad, int act,
+    int split, hipStream_t stream)
 {
     switch (act) {
-        case 0: launch_fused_bf16_nt<NOUT, 0>(masksB, diffB, base, wbg, ey, B, S, Npad, split, stream); break;
-        case 1: launch_fused_bf16_nt<NOUT, 1>(masksB, diffB, base, wbg, ey, B, S, Npad, split, stream); break;
-        default: launch_fused_bf16_nt<NOUT, 2>(masksB, diffB, base, wbg, ey, B, S, Npad, split, stream); break;
+        case 0: launch_fused_bf16_nt<NOUT, 0>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream); break;
+        case 1: launch_fused_bf16_nt<NOUT, 1>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream); break;
+        default: launch_fused_bf16_nt<NOUT, 2>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream); break;
     }
 }
 
 extern "C" int launch_fused_predict_bf16(
-    const uint16_t* masksB_u, const uint16_t* diffB_u, const float* base,
+    const uint8_t* masksU, const uint16_t* diffB_u, const float* base,
     const float* wbg, float* ey, int B, int S, int M, int Npad, int n_out,
     int act, int split, hipStream_t stream)
 {
     if (M > 32 || Npad % 16 != 0 || Npad / 16 > 8 || split < 1 || split > 2)
         return -1;
-    const __bf16* masksB = reinterpret_cast<const __bf16*>(masksB_u);
     const __bf16* diffB = reinterpret_cast<const __bf16*>(diffB_u);
     switch (n_out) {
-        case 1: launch_fused_bf16_act<1>(masksB, diffB, base, wbg, ey, B, S, Npad, act, split, stream); break;
-        case 2: launch_fused_bf16_act<2>(masksB, diffB, base, wbg, ey, B, S, Npad, act, split, stream); break;
-        case 4: launch_fused_bf16_act<4>(masksB, diffB, base, wbg, ey, B, S, Npad, act, split, stream); break;
+        case 1: launch_fused_bf16_act<1>(masksU, diffB, base, wbg, ey, B, S, M, Npad, act, split, stream); break;
+        case 2: launch_fused_bf16_act<2>(masksU, diffB, base, wbg, ey, B, S, M, Npad, act, split, stream); break;
+        case 4: launch_fused_bf16_act<4>(masksU, diffB, base, wbg, ey, B, S, M, Npad, act, split, stream); break;
         default: return -1;
     }
     return 0;
