@@ -438,26 +438,33 @@ __global__ void pack_masks_kernel(
     uint64_t* __restrict__ packed,      // (B, S)
     size_t n_rows, int M)
 {
-    // thread per (row, k<=31); packed bits fall out of a half-wave ballot
-    const size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= n_rows * 32) return;
-    const size_t row = idx >> 5;
-    const int k = idx & 31;
-    const int lane = threadIdx.x & (WAVE - 1);
-    uint8_t v = (k < M) ? (masks[row * M + k] & 1) : 0;
-    uint64_t bal = __ballot(v != 0);
-    if (k == 0)
-        packed[row] = (lane < 32) ? (bal & 0xffffffffull) : (bal >> 32);
+    // one thread per row, bytes gathered with dword loads (a thread-per-bit
+    // ballot variant was load-issue-bound: 32 byte loads per 12-byte row)
+    const size_t row = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (row >= n_rows) return;
+    const uint8_t* src = masks + row * M;
+    uint64_t bits = 0ull;
+    int k = 0;
+    // rows are 4-byte aligned whenever M % 4 == 0 (the common case)
+    if (((size_t)src & 3) == 0) {
+        for (; k + 4 <= M; k += 4) {
+            uint32_t w = *(const uint32_t*)(src + k);
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                bits |= ((uint64_t)((w >> (8 * j)) & 1u)) << (k + j);
+        }
+    }
+    for (; k < M; ++k) bits |= ((uint64_t)(src[k] & 1)) << k;
+    packed[row] = bits;
 }
 
 extern "C" void launch_pack_masks(
     const uint8_t* masks, uint64_t* packed, int B, int S, int M,
     hipStream_t stream)
 {
-    size_t n = (size_t)B * S * 32;
+    size_t n = (size_t)B * S;
     dim3 grid((unsigned)((n + 255) / 256)), block(256);
-    pack_masks_kernel<<<grid, block, 0, stream>>>(
-        masks, packed, (size_t)B * S, M);
+    pack_masks_kernel<<<grid, block, 0, stream>>>(masks, packed, n, M);
 }
 
 template <int NOUT, int ACT, int NT, int SPLIT>  // SPLIT: 1 = hi only, 2 = hi+lo
