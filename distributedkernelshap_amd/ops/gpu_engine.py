@@ -261,20 +261,49 @@ class GpuKernelShap:
         npad = (self.N + 15) // 16 * 16
         if vidx_t is None:
             vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
+        act, oimg = self._act_oimg()
         # pad slots are zeroed once at allocation and never written after
         # (buffer name keyed by m so a different varying count cannot see
         # stale values)
-        diff = self._buf(f"diff{m}", (b, self.n_out, mpad, npad), zeroed=True)
-        self.ext.build_diff_f32(self._x_part(X_dev), self.bg_part, vidx_t, diff)
-        base = t.zeros(self.n_out, npad, device=self.device)
-        base[:, : self.N] = self.baseN.T
+        diff = self._buf(f"diff{m}", (b, oimg, mpad, npad), zeroed=True)
+        self.ext.build_diff_f32(
+            self._x_part_img(X_dev, act), self._bg_part_img(act), vidx_t, diff
+        )
+        base = t.zeros(oimg, npad, device=self.device)
+        base[:, : self.N] = self._base_img(act)
         wbg = t.zeros(npad, device=self.device)
         wbg[: self.N] = self.bg_w
         ey = self._buf("ey", (b, s, self.n_out))
-        self.ext.fused_predict_linear(
-            masksT, diff, base, wbg, ey, self.linear["act"], m
-        )
+        self.ext.fused_predict_linear(masksT, diff, base, wbg, ey, act, m)
         return ey
+
+    def _act_oimg(self):
+        """Binary softmax runs on the logit DIFFERENCE (one operand image,
+        half the MFMAs): act code 3."""
+        act = self.linear["act"]
+        if act == 2 and self.n_out == 2:
+            return 3, 1
+        return act, self.n_out
+
+    def _x_part_img(self, X_dev, act):
+        xp = self._x_part(X_dev)                       # (b, G, o)
+        if act == 3:
+            return (xp[:, :, 1:2] - xp[:, :, 0:1]).contiguous()
+        return xp
+
+    def _bg_part_img(self, act):
+        if act == 3:
+            if not hasattr(self, "_bg_part_d"):
+                self._bg_part_d = (
+                    self.bg_part[:, :, 1:2] - self.bg_part[:, :, 0:1]
+                ).contiguous()
+            return self._bg_part_d
+        return self.bg_part
+
+    def _base_img(self, act):
+        if act == 3:
+            return (self.baseN[:, 1] - self.baseN[:, 0])[None, :]
+        return self.baseN.T
 
     def _ey_fused_bf16(self, masks, X_dev, varying, vidx_t=None, packed=None):
         """bf16 matrix-core predict (predict_dtype 'bf16x2'/'bf16'): one
@@ -288,19 +317,19 @@ class GpuKernelShap:
             self.ext.pack_masks(masks, packed)
         if vidx_t is None:
             vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
+        act, oimg = self._act_oimg()
         diffB = self._buf(
-            f"diffB{m}", (b, split, self.n_out, npad, 40), t.bfloat16,
-            zeroed=True,
+            f"diffB{m}", (b, split, oimg, npad, 40), t.bfloat16, zeroed=True
         )
-        self.ext.build_diff_bf16(self._x_part(X_dev), self.bg_part, vidx_t, diffB)
-        base = t.zeros(self.n_out, npad, device=self.device)
-        base[:, : self.N] = self.baseN.T
+        self.ext.build_diff_bf16(
+            self._x_part_img(X_dev, act), self._bg_part_img(act), vidx_t, diffB
+        )
+        base = t.zeros(oimg, npad, device=self.device)
+        base[:, : self.N] = self._base_img(act)
         wbg = t.zeros(npad, device=self.device)
         wbg[: self.N] = self.bg_w
         ey = self._buf("ey", (b, s, self.n_out))
-        self.ext.fused_predict_bf16(
-            masks, diffB, base, wbg, ey, self.linear["act"]
-        )
+        self.ext.fused_predict_bf16(masks, diffB, base, wbg, ey, act)
         return ey
 
     def _ey_linear_torch(self, masks, X_dev, varying, s_chunk=4096):

@@ -98,10 +98,13 @@ void fused_predict_linear(
     CHECK_DEV(masksT); CHECK_DEV(diff); CHECK_DEV(base); CHECK_DEV(wbg); CHECK_DEV(ey);
     TORCH_CHECK(masksT.dtype() == torch::kFloat32, "masksT must be f32 (B,Mpad,S)");
     int B = masksT.size(0), S = masksT.size(2);
-    int n_out = diff.size(1), Mpad = diff.size(2), Npad = diff.size(3);
+    int Mpad = diff.size(2), Npad = diff.size(3);
+    int n_out = ey.size(2);                    // act 3 uses 1 diff image for 2 outputs
+    int oimg = (act == 3) ? 1 : n_out;
+    TORCH_CHECK(diff.size(1) == oimg, "diff image count vs act");
     TORCH_CHECK(masksT.size(1) == Mpad, "masksT/diff Mpad mismatch");
-    TORCH_CHECK(ey.size(0) == B && ey.size(1) == S && ey.size(2) == n_out, "ey shape");
-    TORCH_CHECK(base.size(0) == n_out && base.size(1) == Npad, "base shape");
+    TORCH_CHECK(ey.size(0) == B && ey.size(1) == S, "ey shape");
+    TORCH_CHECK(base.size(0) == oimg && base.size(1) == Npad, "base shape");
     TORCH_CHECK(wbg.size(0) == Npad, "wbg shape");
     int rc = launch_fused_predict_linear(
         masksT.data_ptr<float>(), diff.data_ptr<float>(), base.data_ptr<float>(),
@@ -164,9 +167,11 @@ void fused_predict_bf16(
     CHECK_DEV(masks); CHECK_DEV(diffB); CHECK_DEV(base); CHECK_DEV(wbg); CHECK_DEV(ey);
     TORCH_CHECK(masks.dtype() == torch::kUInt8 && diffB.dtype() == torch::kBFloat16);
     int B = masks.size(0), S = masks.size(1), M = masks.size(2);
-    int split = diffB.size(1), n_out = diffB.size(2), Npad = diffB.size(3);
+    int split = diffB.size(1), Npad = diffB.size(3);
+    int n_out = ey.size(2);                    // act 3 uses 1 diff image for 2 outputs
+    TORCH_CHECK(diffB.size(2) == ((act == 3) ? 1 : n_out), "diffB image count vs act");
     TORCH_CHECK(diffB.size(0) == B && diffB.size(4) == 40, "diffB (B,split,o,Npad,40)");
-    TORCH_CHECK(ey.size(0) == B && ey.size(1) == S && ey.size(2) == n_out, "ey shape");
+    TORCH_CHECK(ey.size(0) == B && ey.size(1) == S, "ey shape");
     int rc = launch_fused_predict_bf16(
         masks.data_ptr<uint8_t>(),
         reinterpret_cast<const uint16_t*>(diffB.data_ptr<at::BFloat16>()),

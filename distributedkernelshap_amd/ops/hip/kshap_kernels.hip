@@ -216,12 +216,15 @@ template <int NOUT, int ACT, int NT>   // NT = Npad/16 col tiles (constexpr
 __global__ __launch_bounds__(256)      // LDS offsets: the kernel was VALU-
 void fused_predict_linear_kernel(      // bound on runtime address math)
     const float* __restrict__ masksT,   // (B, Mpad, S) zero-padded rows k>=M
-    const float* __restrict__ diff,     // (B, NOUT, Mpad, NT*16)
-    const float* __restrict__ base,     // (NOUT, NT*16)
+    const float* __restrict__ diff,     // (B, OIMG, Mpad, NT*16)
+    const float* __restrict__ base,     // (OIMG, NT*16)
     const float* __restrict__ wbg,      // (NT*16)  0 for padding cols
     float* __restrict__ ey,             // (B, S, NOUT)
     int B, int S, int M, int Mpad)
 {
+    // ACT==3: binary softmax from the logit DIFFERENCE — one operand image
+    // (z1 - z0), half the MFMAs, both outputs from sigma(z) in the epilogue
+    constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
     constexpr int NPAD = NT * 16;
     constexpr int NSTRIDE = NPAD + ((16 - (NPAD & 31)) & 31);  // ≡16 mod 32
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
@@ -233,19 +236,19 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
     const int wave = tid >> 6;           // 0..3
 
     extern __shared__ float lds[];
-    float* diff_lds = lds;                                   // NOUT*Mpad*NSTRIDE
-    float* base_lds = diff_lds + NOUT * Mpad * NSTRIDE;      // NOUT*NPAD
-    float* wbg_lds = base_lds + NOUT * NPAD;                 // NPAD
+    float* diff_lds = lds;                                   // OIMG*Mpad*NSTRIDE
+    float* base_lds = diff_lds + OIMG * Mpad * NSTRIDE;      // OIMG*NPAD
+    float* wbg_lds = base_lds + OIMG * NPAD;                 // NPAD
 
     // ---- stage diff / base / wbg once; the 8 s-subtiles reuse them --------
     const float* msrc = masksT + (size_t)b * Mpad * S;
-    const float* dsrc = diff + (size_t)b * NOUT * Mpad * NPAD;
-    for (int idx = tid; idx < NOUT * Mpad * NPAD; idx += 256) {
+    const float* dsrc = diff + (size_t)b * OIMG * Mpad * NPAD;
+    for (int idx = tid; idx < OIMG * Mpad * NPAD; idx += 256) {
         int ok = idx / NPAD;             // o * Mpad + k
         int n = idx % NPAD;
         diff_lds[ok * NSTRIDE + n] = dsrc[idx];
     }
-    for (int idx = tid; idx < NOUT * NPAD; idx += 256) base_lds[idx] = base[idx];
+    for (int idx = tid; idx < OIMG * NPAD; idx += 256) base_lds[idx] = base[idx];
     for (int idx = tid; idx < NPAD; idx += 256) wbg_lds[idx] = wbg[idx];
     __syncthreads();
 
@@ -282,11 +285,11 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
         const int CTN = half == 0 ? NTH : NT - NTH;
         if (CTN <= 0) continue;
 
-        f32x4 acc[NTH][NOUT];
+        f32x4 acc[NTH][OIMG];
 #pragma unroll
         for (int ct = 0; ct < NTH; ++ct)
 #pragma unroll
-            for (int o = 0; o < NOUT; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
+            for (int o = 0; o < OIMG; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
 
         for (int ks = 0; ks < Mpad; ks += 4) {
             float a = svalid ? mlane[(size_t)ks * S + ssub0] : 0.0f;
@@ -294,7 +297,7 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
             for (int ct = 0; ct < NTH; ++ct) {
                 if (ct < CTN) {
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o) {
+                    for (int o = 0; o < OIMG; ++o) {
                         float bv = dbase[(o * Mpad + ks) * NSTRIDE + (CT0 + ct) * 16];
                         acc[ct][o] =
                             __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[ct][o], 0, 0, 0);
@@ -312,30 +315,37 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
             float wn = wbg_lds[n];
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                float z[NOUT];
+                float z[OIMG];
 #pragma unroll
-                for (int o = 0; o < NOUT; ++o) z[o] = acc[ct][o][r] + base_lds[o * NPAD + n];
-                if (ACT == 1) {
+                for (int o = 0; o < OIMG; ++o) z[o] = acc[ct][o][r] + base_lds[o * NPAD + n];
+                float zz[NOUT];
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o) z[o] = 1.0f / (1.0f + __expf(-z[o]));
+                for (int o = 0; o < NOUT && o < OIMG; ++o) zz[o] = z[o];
+                if (ACT == 3) {
+                    float p1 = 1.0f / (1.0f + __expf(-z[0]));
+                    zz[0] = 1.0f - p1;
+                    zz[NOUT - 1] = p1;
+                } else if (ACT == 1) {
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) zz[o] = 1.0f / (1.0f + __expf(-z[o]));
                 } else if (ACT == 2 && NOUT == 2) {
                     // binary softmax = one sigmoid: p1 = 1/(1+exp(z0-z1))
                     float p1 = 1.0f / (1.0f + __expf(z[0] - z[1]));
-                    z[0] = 1.0f - p1;
-                    z[1] = p1;
+                    zz[0] = 1.0f - p1;
+                    zz[1] = p1;
                 } else if (ACT == 2) {
                     float mx = z[0];
 #pragma unroll
                     for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
                     float sum = 0.0f;
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o) { z[o] = __expf(z[o] - mx); sum += z[o]; }
+                    for (int o = 0; o < NOUT; ++o) { zz[o] = __expf(z[o] - mx); sum += zz[o]; }
                     float inv = 1.0f / sum;
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o) z[o] *= inv;
+                    for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
                 }
 #pragma unroll
-                for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * z[o];
+                for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * zz[o];
             }
         }
         }  // half loop
@@ -373,7 +383,8 @@ static void launch_fused_nt(
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     dim3 grid(B * n_stiles), block(256);
     const int NSTRIDE = Npad + ((16 - (Npad & 31)) & 31);
-    size_t lds = (size_t)(NOUT * Mpad * NSTRIDE + NOUT * Npad + Npad) * 4;
+    const int oimg = (ACT == 3) ? 1 : NOUT;
+    size_t lds = (size_t)(oimg * Mpad * NSTRIDE + oimg * Npad + Npad) * 4;
 #define KSHAP_CASE(NTV) \
     case NTV: \
         fused_predict_linear_kernel<NOUT, ACT, NTV><<<grid, block, lds, stream>>>( \
@@ -397,6 +408,10 @@ static void launch_fused_act(
             break;
         case 1:
             launch_fused_nt<NOUT, 1>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
+            break;
+        case 3:
+            if (NOUT == 2)
+                launch_fused_nt<NOUT, 3>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
         default:
             launch_fused_nt<NOUT, 2>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
@@ -477,6 +492,7 @@ void fused_predict_bf16_kernel(
     float* __restrict__ ey,             // (B, S, NOUT)
     int B, int S, int M)
 {
+    constexpr int OIMG = (ACT == 3) ? 1 : NOUT;   // ACT 3: logit-difference
     constexpr int NPAD = NT * 16;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     const int b = blockIdx.x / n_stiles;
@@ -487,14 +503,14 @@ void fused_predict_bf16_kernel(
     const int wave = tid >> 6;
 
     extern __shared__ float lds[];
-    __bf16* diff_lds = (__bf16*)lds;                 // SPLIT*NOUT*NPAD*KSTRIDE_BF
-    float* base_lds = lds + (SPLIT * NOUT * NPAD * KSTRIDE_BF + 1) / 2;
-    float* wbg_lds = base_lds + NOUT * NPAD;
+    __bf16* diff_lds = (__bf16*)lds;                 // SPLIT*OIMG*NPAD*KSTRIDE_BF
+    float* base_lds = lds + (SPLIT * OIMG * NPAD * KSTRIDE_BF + 1) / 2;
+    float* wbg_lds = base_lds + OIMG * NPAD;
 
-    const __bf16* dsrc = diffB + (size_t)b * SPLIT * NOUT * NPAD * KSTRIDE_BF;
-    for (int idx = tid; idx < SPLIT * NOUT * NPAD * KSTRIDE_BF / 8; idx += 256)
+    const __bf16* dsrc = diffB + (size_t)b * SPLIT * OIMG * NPAD * KSTRIDE_BF;
+    for (int idx = tid; idx < SPLIT * OIMG * NPAD * KSTRIDE_BF / 8; idx += 256)
         ((bf16x8*)diff_lds)[idx] = ((const bf16x8*)dsrc)[idx];
-    for (int idx = tid; idx < NOUT * NPAD; idx += 256) base_lds[idx] = base[idx];
+    for (int idx = tid; idx < OIMG * NPAD; idx += 256) base_lds[idx] = base[idx];
     for (int idx = tid; idx < NPAD; idx += 256) wbg_lds[idx] = wbg[idx];
     __syncthreads();
 
@@ -530,14 +546,14 @@ void fused_predict_bf16_kernel(
 
 #pragma unroll
         for (int ct = 0; ct < NT; ++ct) {
-            f32x4 acc[NOUT];
+            f32x4 acc[OIMG];
 #pragma unroll
-            for (int o = 0; o < NOUT; ++o) {
+            for (int o = 0; o < OIMG; ++o) {
                 acc[o] = (f32x4){0, 0, 0, 0};
 #pragma unroll
                 for (int sp = 0; sp < SPLIT; ++sp) {
                     bf16x8 bv = *(const bf16x8*)(
-                        dlane + ((size_t)(sp * NOUT + o) * NPAD + ct * 16) * KSTRIDE_BF);
+                        dlane + ((size_t)(sp * OIMG + o) * NPAD + ct * 16) * KSTRIDE_BF);
                     acc[o] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a, bv, acc[o], 0, 0, 0);
                 }
@@ -546,29 +562,36 @@ void fused_predict_bf16_kernel(
             float wn = wbg_lds[n];
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                float z[NOUT];
+                float z[OIMG];
 #pragma unroll
-                for (int o = 0; o < NOUT; ++o) z[o] = acc[o][r] + base_lds[o * NPAD + n];
-                if (ACT == 1) {
+                for (int o = 0; o < OIMG; ++o) z[o] = acc[o][r] + base_lds[o * NPAD + n];
+                float zz[NOUT];
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o) z[o] = 1.0f / (1.0f + __expf(-z[o]));
+                for (int o = 0; o < NOUT && o < OIMG; ++o) zz[o] = z[o];
+                if (ACT == 3) {
+                    float p1 = 1.0f / (1.0f + __expf(-z[0]));
+                    zz[0] = 1.0f - p1;
+                    zz[NOUT - 1] = p1;
+                } else if (ACT == 1) {
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) zz[o] = 1.0f / (1.0f + __expf(-z[o]));
                 } else if (ACT == 2 && NOUT == 2) {
                     float p1 = 1.0f / (1.0f + __expf(z[0] - z[1]));
-                    z[0] = 1.0f - p1;
-                    z[1] = p1;
+                    zz[0] = 1.0f - p1;
+                    zz[1] = p1;
                 } else if (ACT == 2) {
                     float mx = z[0];
 #pragma unroll
                     for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
                     float sum = 0.0f;
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o) { z[o] = __expf(z[o] - mx); sum += z[o]; }
+                    for (int o = 0; o < NOUT; ++o) { zz[o] = __expf(z[o] - mx); sum += zz[o]; }
                     float inv = 1.0f / sum;
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o) z[o] *= inv;
+                    for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
                 }
 #pragma unroll
-                for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * z[o];
+                for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * zz[o];
             }
         }
 #pragma unroll
@@ -604,8 +627,9 @@ static void launch_fused_bf16_nt(
 {
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     dim3 grid(B * n_stiles), block(256);
-    size_t lds = ((size_t)split * NOUT * Npad * KSTRIDE_BF * 2 + 2)
-                 + (size_t)(NOUT * Npad + Npad) * 4;
+    const int oimg = (ACT == 3) ? 1 : NOUT;
+    size_t lds = ((size_t)split * oimg * Npad * KSTRIDE_BF * 2 + 2)
+                 + (size_t)(oimg * Npad + Npad) * 4;
     lds = (lds + 3) / 4 * 4 + 4;
 #define KSHAP_BF_CASE(NTV, SPL) \
     if (Npad / 16 == NTV && split == SPL) { \
@@ -629,6 +653,10 @@ static void launch_fused_bf16_act(
     switch (act) {
         case 0: launch_fused_bf16_nt<NOUT, 0>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream); break;
         case 1: launch_fused_bf16_nt<NOUT, 1>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream); break;
+        case 3:
+            if (NOUT == 2)
+                launch_fused_bf16_nt<NOUT, 3>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream);
+            break;
         default: launch_fused_bf16_nt<NOUT, 2>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream); break;
     }
 }
