@@ -410,7 +410,7 @@ static void launch_fused_act(
             launch_fused_nt<NOUT, 1>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
         case 3:
-            if (NOUT == 2)
+            if constexpr (NOUT == 2)
                 launch_fused_nt<NOUT, 3>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
         default:
@@ -654,7 +654,7 @@ static void launch_fused_bf16_act(
         case 0: launch_fused_bf16_nt<NOUT, 0>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream); break;
         case 1: launch_fused_bf16_nt<NOUT, 1>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream); break;
         case 3:
-            if (NOUT == 2)
+            if constexpr (NOUT == 2)
                 launch_fused_bf16_nt<NOUT, 3>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream);
             break;
         default: launch_fused_bf16_nt<NOUT, 2>(masksU, diffB, base, wbg, ey, B, S, M, Npad, split, stream); break;
