@@ -273,9 +273,12 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
         // L2-resident) A stream but halve the accumulator AGPR footprint,
         // buying occupancy (93 VGPR + 56 AGPR -> occ 3 at NT=7 otherwise)
         constexpr int NTH = (NT + 1) / 2;
-        float partial[NOUT][4];
+        // ACT 3 accumulates only p1; ey0 = 1 - ey1 at the write (weights sum
+        // to 1 over the real background columns)
+        constexpr int NACC = (ACT == 3) ? 1 : NOUT;
+        float partial[NACC][4];
 #pragma unroll
-        for (int o = 0; o < NOUT; ++o)
+        for (int o = 0; o < NACC; ++o)
 #pragma unroll
             for (int r = 0; r < 4; ++r) partial[o][r] = 0.0f;
 
@@ -318,13 +321,9 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
                 float z[OIMG];
 #pragma unroll
                 for (int o = 0; o < OIMG; ++o) z[o] = acc[ct][o][r] + base_lds[o * NPAD + n];
-                float zz[NOUT];
-#pragma unroll
-                for (int o = 0; o < NOUT && o < OIMG; ++o) zz[o] = z[o];
+                float zz[NACC];
                 if (ACT == 3) {
-                    float p1 = 1.0f / (1.0f + __expf(-z[0]));
-                    zz[0] = 1.0f - p1;
-                    zz[NOUT - 1] = p1;
+                    zz[0] = 1.0f / (1.0f + __expf(-z[0]));   // p1 only
                 } else if (ACT == 1) {
 #pragma unroll
                     for (int o = 0; o < NOUT; ++o) zz[o] = 1.0f / (1.0f + __expf(-z[o]));
@@ -343,15 +342,18 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
                     float inv = 1.0f / sum;
 #pragma unroll
                     for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
+                } else {
+#pragma unroll
+                    for (int o = 0; o < NACC; ++o) zz[o] = z[o];
                 }
 #pragma unroll
-                for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * zz[o];
+                for (int o = 0; o < NACC; ++o) partial[o][r] += wn * zz[o];
             }
         }
         }  // half loop
         // reduce over the 16 lanes of each row group (xor bits 0-3 in-group)
 #pragma unroll
-        for (int o = 0; o < NOUT; ++o)
+        for (int o = 0; o < NACC; ++o)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 float v = partial[o][r];
@@ -366,9 +368,15 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
             for (int r = 0; r < 4; ++r) {
                 int ss = ssub0 + swave + akcol * 4 + r;
                 if (ss < S) {
+                    if (ACT == 3) {
+                        float p1 = partial[0][r];
+                        ey[((size_t)b * S + ss) * NOUT + 0] = 1.0f - p1;
+                        ey[((size_t)b * S + ss) * NOUT + NOUT - 1] = p1;
+                    } else {
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o)
-                        ey[((size_t)b * S + ss) * NOUT + o] = partial[o][r];
+                        for (int o = 0; o < NACC; ++o)
+                            ey[((size_t)b * S + ss) * NOUT + o] = partial[o][r];
+                    }
                 }
             }
         }
@@ -538,9 +546,10 @@ void fused_predict_bf16_kernel(
                 (svalid && k < M) ? (mrow[k] & 1) : 0);
         }
 
-        float partial[NOUT][4];
+        constexpr int NACC = (ACT == 3) ? 1 : NOUT;
+        float partial[NACC][4];
 #pragma unroll
-        for (int o = 0; o < NOUT; ++o)
+        for (int o = 0; o < NACC; ++o)
 #pragma unroll
             for (int r = 0; r < 4; ++r) partial[o][r] = 0.0f;
 
@@ -565,13 +574,9 @@ void fused_predict_bf16_kernel(
                 float z[OIMG];
 #pragma unroll
                 for (int o = 0; o < OIMG; ++o) z[o] = acc[o][r] + base_lds[o * NPAD + n];
-                float zz[NOUT];
-#pragma unroll
-                for (int o = 0; o < NOUT && o < OIMG; ++o) zz[o] = z[o];
+                float zz[NACC];
                 if (ACT == 3) {
-                    float p1 = 1.0f / (1.0f + __expf(-z[0]));
-                    zz[0] = 1.0f - p1;
-                    zz[NOUT - 1] = p1;
+                    zz[0] = 1.0f / (1.0f + __expf(-z[0]));   // p1 only
                 } else if (ACT == 1) {
 #pragma unroll
                     for (int o = 0; o < NOUT; ++o) zz[o] = 1.0f / (1.0f + __expf(-z[o]));
@@ -589,13 +594,16 @@ void fused_predict_bf16_kernel(
                     float inv = 1.0f / sum;
 #pragma unroll
                     for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
+                } else {
+#pragma unroll
+                    for (int o = 0; o < NACC; ++o) zz[o] = z[o];
                 }
 #pragma unroll
-                for (int o = 0; o < NOUT; ++o) partial[o][r] += wn * zz[o];
+                for (int o = 0; o < NACC; ++o) partial[o][r] += wn * zz[o];
             }
         }
 #pragma unroll
-        for (int o = 0; o < NOUT; ++o)
+        for (int o = 0; o < NACC; ++o)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 float v = partial[o][r];
@@ -610,9 +618,15 @@ void fused_predict_bf16_kernel(
             for (int r = 0; r < 4; ++r) {
                 int ss = ssub0 + swave + akb * 4 + r;
                 if (ss < S) {
+                    if (ACT == 3) {
+                        float p1 = partial[0][r];
+                        ey[((size_t)b * S + ss) * NOUT + 0] = 1.0f - p1;
+                        ey[((size_t)b * S + ss) * NOUT + NOUT - 1] = p1;
+                    } else {
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o)
-                        ey[((size_t)b * S + ss) * NOUT + o] = partial[o][r];
+                        for (int o = 0; o < NACC; ++o)
+                            ey[((size_t)b * S + ss) * NOUT + o] = partial[o][r];
+                    }
                 }
             }
         }
