@@ -1028,6 +1028,7 @@ void wls_solve_mfma_kernel(
     __shared__ float tot_s[WLS_MAX_NOUT];
 
     if (tid < n_out) tot_s[tid] = total[(size_t)b * n_out + tid];
+    __syncthreads();                     // tot_s visible to every wave
 
     const uint64_t* pbase = packed + (size_t)b * S;
     const float* kwb = kw + (size_t)b * S;
@@ -1040,7 +1041,9 @@ void wls_solve_mfma_kernel(
 
     for (int c0 = 0; c0 < S; c0 += WLS_CHUNK) {
         const int clen = min(WLS_CHUNK, S - c0);
-        __syncthreads();
+        // thread tid = wv*64 + lane stages exactly the sample its own wave
+        // consumes below, so chunk staging needs NO barriers — an lgkmcnt
+        // wait makes the wave's LDS writes visible to its own lanes
         if (tid < clen) {
             uint64_t bits = pbase[c0 + tid];
             pk[tid] = bits;
@@ -1049,7 +1052,7 @@ void wls_solve_mfma_kernel(
             for (int o = 0; o < n_out; ++o)
                 eych[tid][o] = eyb[(size_t)(c0 + tid) * n_out + o] - mlast * tot_s[o];
         }
-        __syncthreads();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         // wave wv covers samples [wv*64, wv*64+64) of the chunk, 4 per step
         const int base = wv * 64;
         for (int ks = 0; ks < 64; ks += 4) {
