@@ -60,7 +60,10 @@ def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--workers", type=int, default=-1,
                         help="-1 = sequential baseline (no pool)")
-    parser.add_argument("--batch", nargs="+", type=int, default=[1, 5, 10])
+    parser.add_argument("-b", "--batch", nargs="+", type=int, default=[1, 5, 10])
+    parser.add_argument("--benchmark", type=int, default=1, choices=[0, 1],
+                        help="0 = single run of the first batch size "
+                             "(reference ray_pool.py flag parity)")
     parser.add_argument("--nruns", type=int, default=5)
     parser.add_argument("--instances", type=int, default=2560)
     parser.add_argument("--actor-cpu-fraction", type=float, default=1.0)
@@ -71,6 +74,9 @@ def main():
                         help="TOML config (distributedkernelshap_amd.config) "
                              "overriding bench/engine defaults")
     args = parser.parse_args()
+    if not args.benchmark:
+        args.nruns = 1
+        args.batch = args.batch[:1]
     if args.config_file:
         from distributedkernelshap_amd.config import Config
 

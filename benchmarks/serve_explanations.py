@@ -84,6 +84,8 @@ def main():
                              "one GPU replica already saturates a device")
     parser.add_argument("--max-batch-size", type=int, default=64)
     parser.add_argument("--instances", type=int, default=2560)
+    parser.add_argument("--benchmark", type=int, default=1, choices=[0, 1],
+                        help="0 = single run (reference flag parity)")
     parser.add_argument("--nruns", type=int, default=3)
     parser.add_argument("--concurrency", type=int, default=32)
     parser.add_argument("--batch-mode", default="ray", choices=["ray", "default"],
@@ -97,6 +99,9 @@ def main():
     parser.add_argument("--assets-dir", default="assets")
     parser.add_argument("--results-dir", default="results")
     args = parser.parse_args()
+
+    if not args.benchmark:
+        args.nruns = 1
 
     from distributedkernelshap_amd.utils import get_filename, load_data
 

@@ -56,3 +56,15 @@ def test_analysis_cli(tmp_path):
     r = _run(["benchmarks/analysis.py", "--results-dir", results])
     assert r.returncode == 0
     assert "pool" in r.stdout and "4" in r.stdout
+
+
+def test_serve_cli_tiny(tmp_path):
+    assets = str(tmp_path / "assets")
+    results = str(tmp_path / "results")
+    r = _run([
+        "benchmarks/serve_explanations.py", "--instances", "6", "--nruns", "1",
+        "--benchmark", "0", "--concurrency", "4", "--port", "8877",
+        "--assets-dir", assets, "--results-dir", results,
+    ], timeout=900)
+    assert r.returncode == 0, (r.stderr[-800:], r.stdout[-300:])
+    assert os.path.exists(os.path.join(results, "ray_replicas_1_maxbatch_64.pkl"))
