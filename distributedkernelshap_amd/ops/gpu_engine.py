@@ -565,13 +565,32 @@ class GpuKernelShap:
         phi_full = t.zeros(b, self.n_groups, self.n_out, device=self.device)
 
         # bucket instances by varying-group pattern (benchmark case: 1 bucket)
-        vmat = self._varying_matrix(X_dev)              # (B, G) bool, host
-        timer.mark("varying")
-        if self.n_groups <= 64:
-            keys = vmat @ (1 << np.arange(self.n_groups, dtype=np.uint64))
-            _, first, inverse = np.unique(keys, return_index=True, return_inverse=True)
-            uniq = vmat[first]
+        if self.n_groups <= 62:
+            # fast path: pack each instance's pattern into an int64 key on
+            # device and transfer only (all_same, key0) — 16 bytes instead of
+            # the (B, G) matrix (the D2H sync dominated this stage)
+            gbool = self._varying_matrix_dev(X_dev)     # (B, G) bool, device
+            keys = (gbool.long() << t.arange(
+                self.n_groups, device=self.device)).sum(dim=1)
+            probe = t.stack([(keys == keys[0]).all().long(), keys[0]]).cpu()
+            timer.mark("varying")
+            if bool(probe[0]):
+                key0 = int(probe[1])
+                uniq = np.array(
+                    [[(key0 >> g) & 1 for g in range(self.n_groups)]],
+                    dtype=bool,
+                )
+                inverse = np.zeros(b, dtype=np.int64)
+            else:
+                vmat = gbool.cpu().numpy()
+                ks = vmat @ (1 << np.arange(self.n_groups, dtype=np.uint64))
+                _, first, inverse = np.unique(
+                    ks, return_index=True, return_inverse=True
+                )
+                uniq = vmat[first]
         else:
+            vmat = self._varying_matrix_dev(X_dev).cpu().numpy()
+            timer.mark("varying")
             uniq, inverse = np.unique(vmat, axis=0, return_inverse=True)
         timer.mark("bucket")
 
@@ -684,8 +703,8 @@ class GpuKernelShap:
 
     # ------------------------------------------------------------------ #
 
-    def _varying_matrix(self, X_dev) -> np.ndarray:
-        """K1: per-instance varying-group booleans.
+    def _varying_matrix_dev(self, X_dev):
+        """K1: per-instance varying-group booleans (device tensor).
 
         A column varies iff some background value differs from x there; all
         values lie in [bg_min, bg_max], so it suffices to test the extremes
@@ -701,7 +720,7 @@ class GpuKernelShap:
         coldiff = (~(close_min & close_max)).to(t.int32)  # (B, D)
         gcount = t.zeros(b, self.n_groups, dtype=t.int32, device=self.device)
         gcount.index_add_(1, self.col_group, coldiff)
-        return (gcount > 0).cpu().numpy()
+        return gcount > 0
 
     def _l1_active(self, plan, l1_reg) -> bool:
         m = plan.m
