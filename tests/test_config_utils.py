@@ -141,3 +141,20 @@ n_workers = 2
     assert os.path.exists(
         os.path.join(str(tmp_path), "res", "ray_workers_2_bsize_3_actorfr_1.0.pkl")
     )
+
+
+def test_kernel_config_on_cpu_engine():
+    """predict_dtype/wls_mode are GPU dispatch knobs; a CPU engine accepts
+    them without effect."""
+    from distributedkernelshap_amd.config import KernelConfig
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor
+
+    rng = np.random.default_rng(0)
+    pred = LinearPredictor.random(6, 2)
+    eng = KernelShapEngine(
+        pred, rng.normal(size=(10, 6)), link="logit", device="cpu",
+        kernels=KernelConfig(predict_dtype="bf16x2", wls_mode="torch"),
+    )
+    sv = eng.shap_values(rng.normal(size=(2, 6)))
+    assert sv[0].shape == (2, 6)
