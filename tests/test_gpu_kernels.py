@@ -596,3 +596,36 @@ def test_bf16_single_local_accuracy():
     eng32, _, _ = _bf16_engine("fp32")
     sv32 = eng32.shap_values(data.X)
     assert np.abs(sv[0] - sv32[0]).max() < 0.1
+
+
+def test_build_diff_and_pack_kernels(ext):
+    """Operand-layout builders vs plain torch references."""
+    g = torch.Generator(device="cuda").manual_seed(17)
+    b, G, O, n, m = 3, 9, 2, 20, 5
+    mpad, npad = 8, 32
+    xp = torch.randn(b, G, O, generator=g, device="cuda")
+    bgp = torch.randn(n, G, O, generator=g, device="cuda")
+    vidx = torch.tensor([0, 2, 3, 7, 8], dtype=torch.int64, device="cuda")
+    out = torch.zeros(b, O, mpad, npad, device="cuda")
+    ext.build_diff_f32(xp, bgp, vidx, out)
+    ref = (xp[:, vidx].permute(0, 2, 1)[:, :, :, None]
+           - bgp[:, vidx].permute(2, 1, 0)[None])          # (b, O, m, n)
+    assert torch.allclose(out[:, :, :m, :n], ref, atol=1e-6)
+    assert torch.all(out[:, :, m:, :] == 0) and torch.all(out[:, :, :, n:] == 0)
+
+    outb = torch.zeros(b, 2, O, npad, 40, dtype=torch.bfloat16, device="cuda")
+    ext.build_diff_bf16(xp, bgp, vidx, outb)
+    recon = outb[:, 0].float() + outb[:, 1].float()        # hi + lo
+    # (b, O, n, k) vs ref (b, O, k, n)
+    assert torch.allclose(
+        recon[:, :, :n, :m], ref.permute(0, 1, 3, 2), atol=1e-5
+    )
+
+    masks = (torch.rand(2, 30, 12, generator=g, device="cuda") > 0.5).to(torch.uint8)
+    packed = torch.empty(2, 30, dtype=torch.int64, device="cuda")
+    ext.pack_masks(masks, packed)
+    mh, ph = masks.cpu().numpy(), packed.cpu().numpy().astype(np.uint64)
+    for bi in range(2):
+        for si in range(30):
+            bits = sum(int(mh[bi, si, k]) << k for k in range(12))
+            assert ph[bi, si] == bits

@@ -1,5 +1,6 @@
 """WLS solver tests (CPU oracle of HIP kernel K7)."""
 import numpy as np
+import pytest
 
 from distributedkernelshap_amd.core.sampler import plan_coalitions, sample_masks
 from distributedkernelshap_amd.core.solver import solve_wls
@@ -67,3 +68,23 @@ def test_single_feature():
         np.array([0.7]),
     )
     assert np.allclose(phi, [[0.7]])
+
+
+def test_singular_gram_falls_back_to_lstsq():
+    """Duplicate mask columns make the Gram singular; the solver must not
+    raise (np.linalg fallback, mirroring shap's failure modes)."""
+    s, m = 50, 4
+    rng = np.random.default_rng(0)
+    masks = (rng.random((s, m)) > 0.5).astype(np.uint8)
+    masks[:, 1] = masks[:, 0]  # identical columns -> singular normal matrix
+    kw = np.ones(s)
+    ey = rng.normal(size=(s, 1))
+    total = np.array([1.0])
+    phi = solve_wls(masks, kw, ey, total)
+    assert np.all(np.isfinite(phi))
+    assert np.allclose(phi.sum(axis=0), total)
+
+
+def test_plan_rejects_m_below_two():
+    with pytest.raises(ValueError):
+        plan_coalitions(1)
