@@ -206,6 +206,13 @@ extern "C" void launch_fill_random_masks(
 // Template params: NOUT in {1,2,4}; ACT 0=none 1=sigmoid 2=softmax.
 // ------------------------------------------------------------------------- //
 
+
+// 1-ulp hardware reciprocal (v_rcp_f32): the epilogue sigmoids do not need
+// the IEEE-correct 5-instruction division sequence hipcc emits for `/`
+__device__ __forceinline__ float fast_rcp(float x) {
+    return __builtin_amdgcn_rcpf(x);
+}
+
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define MAX_MPAD 64        // fused path supports up to 64 varying groups
@@ -323,13 +330,13 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
                 for (int o = 0; o < OIMG; ++o) z[o] = acc[ct][o][r] + base_lds[o * NPAD + n];
                 float zz[NACC];
                 if (ACT == 3) {
-                    zz[0] = 1.0f / (1.0f + __expf(-z[0]));   // p1 only
+                    zz[0] = fast_rcp(1.0f + __expf(-z[0]));   // p1 only
                 } else if (ACT == 1) {
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o) zz[o] = 1.0f / (1.0f + __expf(-z[o]));
+                    for (int o = 0; o < NOUT; ++o) zz[o] = fast_rcp(1.0f + __expf(-z[o]));
                 } else if (ACT == 2 && NOUT == 2) {
                     // binary softmax = one sigmoid: p1 = 1/(1+exp(z0-z1))
-                    float p1 = 1.0f / (1.0f + __expf(z[0] - z[1]));
+                    float p1 = fast_rcp(1.0f + __expf(z[0] - z[1]));
                     zz[0] = 1.0f - p1;
                     zz[1] = p1;
                 } else if (ACT == 2) {
@@ -339,7 +346,7 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
                     float sum = 0.0f;
 #pragma unroll
                     for (int o = 0; o < NOUT; ++o) { zz[o] = __expf(z[o] - mx); sum += zz[o]; }
-                    float inv = 1.0f / sum;
+                    float inv = fast_rcp(sum);
 #pragma unroll
                     for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
                 } else {
@@ -576,12 +583,12 @@ void fused_predict_bf16_kernel(
                 for (int o = 0; o < OIMG; ++o) z[o] = acc[o][r] + base_lds[o * NPAD + n];
                 float zz[NACC];
                 if (ACT == 3) {
-                    zz[0] = 1.0f / (1.0f + __expf(-z[0]));   // p1 only
+                    zz[0] = fast_rcp(1.0f + __expf(-z[0]));   // p1 only
                 } else if (ACT == 1) {
 #pragma unroll
-                    for (int o = 0; o < NOUT; ++o) zz[o] = 1.0f / (1.0f + __expf(-z[o]));
+                    for (int o = 0; o < NOUT; ++o) zz[o] = fast_rcp(1.0f + __expf(-z[o]));
                 } else if (ACT == 2 && NOUT == 2) {
-                    float p1 = 1.0f / (1.0f + __expf(z[0] - z[1]));
+                    float p1 = fast_rcp(1.0f + __expf(z[0] - z[1]));
                     zz[0] = 1.0f - p1;
                     zz[1] = p1;
                 } else if (ACT == 2) {
@@ -591,7 +598,7 @@ void fused_predict_bf16_kernel(
                     float sum = 0.0f;
 #pragma unroll
                     for (int o = 0; o < NOUT; ++o) { zz[o] = __expf(z[o] - mx); sum += zz[o]; }
-                    float inv = 1.0f / sum;
+                    float inv = fast_rcp(sum);
 #pragma unroll
                     for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
                 } else {
