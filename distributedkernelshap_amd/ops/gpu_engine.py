@@ -253,11 +253,12 @@ class GpuKernelShap:
         bgv = self.bg_part[:, vidx]             # (N, m, o)
         return xv.permute(0, 2, 1)[:, :, :, None] - bgv.permute(2, 1, 0)[None]
 
-    def _ey_fused_linear(self, masks, masksT, X_dev, varying, vidx_t=None):
-        """K3-K6 fused MFMA path (linear predictor, Mpad<=64, Npad<=128)."""
+    def _ey_fused_linear(self, masks, X_dev, varying, vidx_t=None):
+        """K3-K6 fused MFMA path (linear predictor, Mpad<=64, Npad<=128).
+        The A operand is converted in-register from the raw u8 masks."""
         t = self.torch
         b, s, m = masks.shape
-        mpad = masksT.shape[1]
+        mpad = max(4, (m + 3) // 4 * 4)
         npad = (self.N + 15) // 16 * 16
         if vidx_t is None:
             vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
@@ -274,7 +275,7 @@ class GpuKernelShap:
         wbg = t.zeros(npad, device=self.device)
         wbg[: self.N] = self.bg_w
         ey = self._buf("ey", (b, s, self.n_out))
-        self.ext.fused_predict_linear(masksT, diff, base, wbg, ey, act, m)
+        self.ext.fused_predict_linear(masks, diff, base, wbg, ey, act)
         return ey
 
     def _act_oimg(self):
@@ -314,7 +315,7 @@ class GpuKernelShap:
         npad = (self.N + 15) // 16 * 16
         split = 2 if self.engine.kernels.predict_dtype == "bf16x2" else 1
         if packed is not None:
-            self.ext.pack_masks(masks, packed)
+            self.ext.pack_masks(masks, packed)   # graph-path convenience
         if vidx_t is None:
             vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
         act, oimg = self._act_oimg()
@@ -429,14 +430,12 @@ class GpuKernelShap:
         masks, kw = self._device_masks(plan, np.arange(b), ids_dev=ids_dev)
         kc = self.engine.kernels
         packed = self._buf("packed", (b, plan.nsamples), t.int64)
+        self.ext.pack_masks(masks, packed)
         if (kc.predict_dtype in ("bf16", "bf16x2") and m <= 32
                 and self.n_out in (1, 2, 4)):
-            ey = self._ey_fused_bf16(masks, X_dev, varying, vidx_t, packed=packed)
+            ey = self._ey_fused_bf16(masks, X_dev, varying, vidx_t, packed=None)
         else:
-            mpad = max(4, (m + 3) // 4 * 4)
-            masksT = self._buf("masksT", (b, mpad, plan.nsamples))
-            self.ext.transpose_masks(masks, masksT, packed)
-            ey = self._ey_fused_linear(masks, masksT, X_dev, varying, vidx_t)
+            ey = self._ey_fused_linear(masks, X_dev, varying, vidx_t)
         if self.link_name == "identity":
             ey_adj = ey.sub_(lfnull[None, None, :])
         else:
@@ -640,28 +639,18 @@ class GpuKernelShap:
                     kc.predict_dtype in ("bf16", "bf16x2")
                     and m <= 32 and npad <= 128 and self.n_out in (1, 2, 4)
                 )
+                if packed is not None:
+                    self.ext.pack_masks(masks, packed)
                 if kc.fused_predict and use_bf16:
-                    ey = self._ey_fused_bf16(masks, sub_X, varying, packed=packed)
+                    ey = self._ey_fused_bf16(masks, sub_X, varying, packed=None)
                 elif (kc.fused_predict and mpad <= 64 and npad <= 128
                         and self.n_out in (1, 2, 4)):
-                    masksT = self._buf("masksT", (len(ids), mpad, plan.nsamples))
-                    self.ext.transpose_masks(masks, masksT, packed)
-                    ey = self._ey_fused_linear(masks, masksT, sub_X, varying)
+                    ey = self._ey_fused_linear(masks, sub_X, varying)
                 else:
-                    if packed is not None:
-                        self.ext.transpose_masks(
-                            masks,
-                            self._buf("masksT", (len(ids), max(4, (m + 3) // 4 * 4), plan.nsamples)),
-                            packed,
-                        )
                     ey = self._ey_linear_torch(masks, sub_X, varying)
             else:
                 if packed is not None:
-                    self.ext.transpose_masks(
-                        masks,
-                        self._buf("masksT", (len(ids), max(4, (m + 3) // 4 * 4), plan.nsamples)),
-                        packed,
-                    )
+                    self.ext.pack_masks(masks, packed)
                 ey = self._ey_torch_module(masks, sub_X, varying)
             timer.mark("predict")
             # in-place link transform: ey is a workspace, not needed afterwards

@@ -11,13 +11,9 @@ extern "C" void launch_fill_random_masks(
     uint32_t seed, const int32_t* inst_ids, hipStream_t stream);
 
 extern "C" int launch_fused_predict_linear(
-    const float* masksT, const float* diff, const float* base, const float* wbg,
+    const uint8_t* masksU, const float* diff, const float* base, const float* wbg,
     float* ey, int B, int S, int M, int Mpad, int Npad, int n_out, int act,
     hipStream_t stream);
-
-extern "C" void launch_transpose_masks(
-    const uint8_t* masks, float* masksT, uint64_t* packed, int B, int S,
-    int M, int Mpad, hipStream_t stream);
 
 extern "C" void launch_synth_chunk(
     const uint8_t* masks, const float* x, const float* bg, const int* col_group,
@@ -73,42 +69,23 @@ void fill_random_masks(
         current_stream());
 }
 
-void transpose_masks(torch::Tensor masks, torch::Tensor masksT,
-                     c10::optional<torch::Tensor> packed) {
-    CHECK_DEV(masks); CHECK_DEV(masksT);
-    TORCH_CHECK(masks.dtype() == torch::kUInt8 && masksT.dtype() == torch::kFloat32);
-    int B = masks.size(0), S = masks.size(1), M = masks.size(2);
-    int Mpad = masksT.size(1);
-    TORCH_CHECK(masksT.size(0) == B && masksT.size(2) == S && Mpad >= M, "masksT shape");
-    uint64_t* pk = nullptr;
-    if (packed.has_value()) {
-        CHECK_DEV(packed.value());
-        TORCH_CHECK(M <= 64 && packed->size(0) == B && packed->size(1) == S,
-                    "packed shape/M");
-        pk = reinterpret_cast<uint64_t*>(packed->data_ptr<int64_t>());
-    }
-    launch_transpose_masks(
-        masks.data_ptr<uint8_t>(), masksT.data_ptr<float>(), pk, B, S, M,
-        Mpad, current_stream());
-}
-
 void fused_predict_linear(
-    torch::Tensor masksT, torch::Tensor diff, torch::Tensor base,
-    torch::Tensor wbg, torch::Tensor ey, int64_t act, int64_t m) {
-    CHECK_DEV(masksT); CHECK_DEV(diff); CHECK_DEV(base); CHECK_DEV(wbg); CHECK_DEV(ey);
-    TORCH_CHECK(masksT.dtype() == torch::kFloat32, "masksT must be f32 (B,Mpad,S)");
-    int B = masksT.size(0), S = masksT.size(2);
+    torch::Tensor masks, torch::Tensor diff, torch::Tensor base,
+    torch::Tensor wbg, torch::Tensor ey, int64_t act) {
+    CHECK_DEV(masks); CHECK_DEV(diff); CHECK_DEV(base); CHECK_DEV(wbg); CHECK_DEV(ey);
+    TORCH_CHECK(masks.dtype() == torch::kUInt8, "masks must be u8 (B,S,M)");
+    int B = masks.size(0), S = masks.size(1), M = masks.size(2);
     int Mpad = diff.size(2), Npad = diff.size(3);
     int n_out = ey.size(2);                    // act 3 uses 1 diff image for 2 outputs
     int oimg = (act == 3) ? 1 : n_out;
     TORCH_CHECK(diff.size(1) == oimg, "diff image count vs act");
-    TORCH_CHECK(masksT.size(1) == Mpad, "masksT/diff Mpad mismatch");
+    TORCH_CHECK(M <= Mpad, "M vs Mpad");
     TORCH_CHECK(ey.size(0) == B && ey.size(1) == S, "ey shape");
     TORCH_CHECK(base.size(0) == oimg && base.size(1) == Npad, "base shape");
     TORCH_CHECK(wbg.size(0) == Npad, "wbg shape");
     int rc = launch_fused_predict_linear(
-        masksT.data_ptr<float>(), diff.data_ptr<float>(), base.data_ptr<float>(),
-        wbg.data_ptr<float>(), ey.data_ptr<float>(), B, S, (int)m, Mpad, Npad,
+        masks.data_ptr<uint8_t>(), diff.data_ptr<float>(), base.data_ptr<float>(),
+        wbg.data_ptr<float>(), ey.data_ptr<float>(), B, S, M, Mpad, Npad,
         n_out, (int)act, current_stream());
     TORCH_CHECK(rc == 0, "fused_predict_linear: unsupported shape (Mpad<=64, Npad%16==0, Npad<=128, n_out in {1,2,4})");
 }
@@ -153,7 +130,7 @@ void pack_masks(torch::Tensor masks, torch::Tensor packed) {
     CHECK_DEV(masks); CHECK_DEV(packed);
     TORCH_CHECK(masks.dtype() == torch::kUInt8, "masks must be u8");
     int B = masks.size(0), S = masks.size(1), M = masks.size(2);
-    TORCH_CHECK(M <= 32, "pack_masks supports M <= 32");
+    TORCH_CHECK(M <= 64, "pack_masks supports M <= 64");
     TORCH_CHECK(packed.size(0) == B && packed.size(1) == S, "packed shape");
     launch_pack_masks(
         masks.data_ptr<uint8_t>(),
@@ -222,11 +199,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_predict_bf16", &fused_predict_bf16,
           "bf16 matrix-core fused predict (single or hi+lo split), A-operand "
           "converted in-register from the raw u8 masks");
-    m.def("transpose_masks", &transpose_masks,
-          "masks (B,S,M) u8 -> (B,Mpad,S) f32 (+ packed u64) for coalesced "
-          "A staging (K2b)",
-          pybind11::arg("masks"), pybind11::arg("masksT"),
-          pybind11::arg("packed") = pybind11::none());
     m.def("synth_chunk", &synth_chunk,
           "masked-background perturbation synthesis tile (K3')");
     m.def("wls_solve", &wls_solve,

@@ -222,7 +222,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 template <int NOUT, int ACT, int NT>   // NT = Npad/16 col tiles (constexpr
 __global__ __launch_bounds__(256)      // LDS offsets: the kernel was VALU-
 void fused_predict_linear_kernel(      // bound on runtime address math)
-    const float* __restrict__ masksT,   // (B, Mpad, S) zero-padded rows k>=M
+    const uint8_t* __restrict__ masksU, // (B, S, M) raw coalition masks
     const float* __restrict__ diff,     // (B, OIMG, Mpad, NT*16)
     const float* __restrict__ base,     // (OIMG, NT*16)
     const float* __restrict__ wbg,      // (NT*16)  0 for padding cols
@@ -248,7 +248,6 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
     float* wbg_lds = base_lds + OIMG * NPAD;                 // NPAD
 
     // ---- stage diff / base / wbg once; the 8 s-subtiles reuse them --------
-    const float* msrc = masksT + (size_t)b * Mpad * S;
     const float* dsrc = diff + (size_t)b * OIMG * Mpad * NPAD;
     for (int idx = tid; idx < OIMG * Mpad * NPAD; idx += 256) {
         int ok = idx / NPAD;             // o * Mpad + k
@@ -264,11 +263,10 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
     const int akcol = lane >> 4;         // k within the 4-wide micro-step
     // per-lane LDS base pointer: all loop offsets are compile-time now
     const float* dbase = diff_lds + akcol * NSTRIDE + arow;
-    // A-operand straight from the transposed global layout: every mask
-    // element is consumed exactly ONCE per launch, so LDS staging (and its
-    // per-sub-tile barriers) was pure overhead; the 16-lane groups read 64 B
-    // consecutive segments (coalesced stream)
-    const float* mlane = msrc + (size_t)akcol * S + swave + arow;
+    // A-operand converted in-register straight from the raw u8 masks (every
+    // element is consumed exactly once per launch; a dedicated f32 transposed
+    // image cost a transpose kernel + 0.5 GB/step of HBM traffic)
+    const uint8_t* mlane = masksU + ((size_t)b * S + swave + arow) * M;
 
     for (int sub = 0; sub < S_TILE / S_SUB; ++sub) {
         const int ssub0 = s0 + sub * S_SUB;
@@ -301,8 +299,10 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
 #pragma unroll
             for (int o = 0; o < OIMG; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
 
+        const uint8_t* mrow = mlane + (size_t)ssub0 * M;
         for (int ks = 0; ks < Mpad; ks += 4) {
-            float a = svalid ? mlane[(size_t)ks * S + ssub0] : 0.0f;
+            const int k = ks + akcol;
+            float a = (svalid && k < M) ? (float)(mrow[k] & 1) : 0.0f;
 #pragma unroll
             for (int ct = 0; ct < NTH; ++ct) {
                 if (ct < CTN) {
@@ -392,7 +392,7 @@ void fused_predict_linear_kernel(      // bound on runtime address math)
 
 template <int NOUT, int ACT>
 static void launch_fused_nt(
-    const float* masksT, const float* diff, const float* base, const float* wbg,
+    const uint8_t* masksU, const float* diff, const float* base, const float* wbg,
     float* ey, int B, int S, int M, int Mpad, int Npad, hipStream_t stream)
 {
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
@@ -403,7 +403,7 @@ static void launch_fused_nt(
 #define KSHAP_CASE(NTV) \
     case NTV: \
         fused_predict_linear_kernel<NOUT, ACT, NTV><<<grid, block, lds, stream>>>( \
-            masksT, diff, base, wbg, ey, B, S, M, Mpad); \
+            masksU, diff, base, wbg, ey, B, S, M, Mpad); \
         break;
     switch (Npad / 16) {
         KSHAP_CASE(1) KSHAP_CASE(2) KSHAP_CASE(3) KSHAP_CASE(4)
@@ -414,36 +414,36 @@ static void launch_fused_nt(
 
 template <int NOUT>
 static void launch_fused_act(
-    const float* masksT, const float* diff, const float* base, const float* wbg,
+    const uint8_t* masksU, const float* diff, const float* base, const float* wbg,
     float* ey, int B, int S, int M, int Mpad, int Npad, int act, hipStream_t stream)
 {
     switch (act) {
         case 0:
-            launch_fused_nt<NOUT, 0>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
+            launch_fused_nt<NOUT, 0>(masksU, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
         case 1:
-            launch_fused_nt<NOUT, 1>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
+            launch_fused_nt<NOUT, 1>(masksU, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
         case 3:
             if constexpr (NOUT == 2)
-                launch_fused_nt<NOUT, 3>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
+                launch_fused_nt<NOUT, 3>(masksU, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
         default:
-            launch_fused_nt<NOUT, 2>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
+            launch_fused_nt<NOUT, 2>(masksU, diff, base, wbg, ey, B, S, M, Mpad, Npad, stream);
             break;
     }
 }
 
 extern "C" int launch_fused_predict_linear(
-    const float* masksT, const float* diff, const float* base, const float* wbg,
+    const uint8_t* masksU, const float* diff, const float* base, const float* wbg,
     float* ey, int B, int S, int M, int Mpad, int Npad, int n_out, int act,
     hipStream_t stream)
 {
     if (Mpad > MAX_MPAD || Npad % 16 != 0 || Npad / 16 > 8) return -1;
     switch (n_out) {
-        case 1: launch_fused_act<1>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
-        case 2: launch_fused_act<2>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
-        case 4: launch_fused_act<4>(masksT, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 1: launch_fused_act<1>(masksU, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 2: launch_fused_act<2>(masksU, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 4: launch_fused_act<4>(masksU, diff, base, wbg, ey, B, S, M, Mpad, Npad, act, stream); break;
         default: return -1;
     }
     return 0;
@@ -964,41 +964,6 @@ void wls_solve_kernel(
     }
 }
 
-
-// ------------------------------------------------------------------------- //
-// K2b: transpose masks (B,S,M) u8 -> (B,Mpad,S) f32 so the predict kernel
-// stages its A-operand with coalesced float loads.
-// ------------------------------------------------------------------------- //
-
-__global__ void transpose_masks_kernel(
-    const uint8_t* __restrict__ masks,  // (B, S, M)
-    float* __restrict__ masksT,         // (B, Mpad, S)
-    uint64_t* __restrict__ packed,      // (B, S) bit i = mask[b,s,i]; or null
-    int B, int S, int M, int Mpad)
-{
-    const int b = blockIdx.y;
-    const int s = blockIdx.x * blockDim.x + threadIdx.x;
-    if (s >= S) return;
-    const uint8_t* row = masks + ((size_t)b * S + s) * M;
-    float* out = masksT + (size_t)b * Mpad * S + s;
-    uint64_t bits = 0ull;
-    for (int k = 0; k < M; ++k) {
-        uint8_t v = row[k] & 1;
-        out[(size_t)k * S] = (float)v;
-        bits |= ((uint64_t)v) << k;
-    }
-    for (int k = M; k < Mpad; ++k) out[(size_t)k * S] = 0.0f;
-    if (packed) packed[(size_t)b * S + s] = bits;
-}
-
-extern "C" void launch_transpose_masks(
-    const uint8_t* masks, float* masksT, uint64_t* packed, int B, int S,
-    int M, int Mpad, hipStream_t stream)
-{
-    dim3 grid((S + 255) / 256, B), block(256);
-    transpose_masks_kernel<<<grid, block, 0, stream>>>(
-        masks, masksT, packed, B, S, M, Mpad);
-}
 
 // ------------------------------------------------------------------------- //
 // K7-MFMA: Gram + rhs build on matrix cores for mm + n_out <= 16.

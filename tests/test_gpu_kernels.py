@@ -115,11 +115,6 @@ def test_fused_predict_linear_vs_torch(ext, act, n_out):
     b, s, m, mpad, npad = 3, 200, 12, 12, 112
     n = 100
     masks = (torch.rand(b, s, m, generator=g, device="cuda") > 0.5).to(torch.uint8)
-    masksT = torch.empty(b, mpad, s, device="cuda")
-    ext.transpose_masks(masks, masksT)
-    # transpose correctness (K2b)
-    assert torch.equal(masksT[:, :m].permute(0, 2, 1), masks.float())
-    assert torch.all(masksT[:, m:] == 0)
     diff = torch.zeros(b, n_out, mpad, npad, device="cuda")
     diff[:, :, :m, :n] = torch.randn(b, n_out, m, n, generator=g, device="cuda")
     base = torch.zeros(n_out, npad, device="cuda")
@@ -127,7 +122,7 @@ def test_fused_predict_linear_vs_torch(ext, act, n_out):
     wbg = torch.zeros(npad, device="cuda")
     wbg[:n] = 1.0 / n
     ey = torch.empty(b, s, n_out, device="cuda")
-    ext.fused_predict_linear(masksT, diff, base, wbg, ey, act, m)
+    ext.fused_predict_linear(masks, diff, base, wbg, ey, act)
     ref = _fused_reference(masks, diff, base, wbg, act)
     assert torch.allclose(ey, ref, atol=2e-5, rtol=1e-4), (
         (ey - ref).abs().max().item()
@@ -140,8 +135,6 @@ def test_fused_predict_nonmultiple_shapes(ext):
     b, s, m, n, n_out = 2, 130, 10, 37, 2
     mpad, npad = 12, 48
     masks = (torch.rand(b, s, m, generator=g, device="cuda") > 0.3).to(torch.uint8)
-    masksT = torch.empty(b, mpad, s, device="cuda")
-    ext.transpose_masks(masks, masksT)
     diff = torch.zeros(b, n_out, mpad, npad, device="cuda")
     diff[:, :, :m, :n] = torch.randn(b, n_out, m, n, generator=g, device="cuda")
     base = torch.zeros(n_out, npad, device="cuda")
@@ -149,7 +142,7 @@ def test_fused_predict_nonmultiple_shapes(ext):
     wbg = torch.zeros(npad, device="cuda")
     wbg[:n] = torch.rand(n, generator=g, device="cuda") + 0.1
     ey = torch.empty(b, s, n_out, device="cuda")
-    ext.fused_predict_linear(masksT, diff, base, wbg, ey, 2, m)
+    ext.fused_predict_linear(masks, diff, base, wbg, ey, 2)
     ref = _fused_reference(masks, diff, base, wbg, 2)
     assert torch.allclose(ey, ref, atol=2e-5, rtol=1e-4)
 
@@ -195,9 +188,8 @@ def test_wls_solve_vs_cpu(ext):
     # generic scalar kernel (no packed masks)
     ext.wls_solve(masks, kw, ey, total, phi)
     # MFMA Gram-build kernel (packed masks; (M-1)+n_out <= 16)
-    masksT = torch.empty(b, 12, s, device="cuda")
     packed = torch.empty(b, s, dtype=torch.int64, device="cuda")
-    ext.transpose_masks(masks, masksT, packed)
+    ext.pack_masks(masks, packed)
     phi_mfma = torch.empty_like(phi)
     ext.wls_solve(masks, kw, ey, total, phi_mfma, packed)
     ph = phi.cpu().numpy()
