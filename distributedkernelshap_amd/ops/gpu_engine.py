@@ -513,8 +513,8 @@ class GpuKernelShap:
         from ..core.sampler import default_nsamples
 
         s = nsamples or default_nsamples(max(2, self.n_groups))
-        # masks u8 + masksT f32 + packed u64 + kwb f32 + ey f32 per instance
-        per_inst = s * (5 * self.n_groups + 8 + 4 + 4 * self.n_out) + (1 << 12)
+        # masks u8 + diff images + packed u64 + kwb f32 + ey f32 per instance
+        per_inst = s * (self.n_groups + 8 + 4 + 4 * self.n_out) + (1 << 14)
         return max(1, self._CHUNK_BYTES // per_inst)
 
     def shap_values(
