@@ -123,6 +123,9 @@ class GpuKernelShap:
         # KSHAP_GRAPH=0 or automatically on capture failure
         self._graphs: dict = {}
         self._graph_hits: dict = {}
+        # speculative replay state: (probe_key0, graph key) of the last
+        # successful single-bucket graph call
+        self._spec = None
         self._graphs_enabled = _os.environ.get("KSHAP_GRAPH", "1") == "1"
         # capture costs ~100 ms: only worth it for recurring shapes (a
         # serving workload with varying batch sizes must stay eager)
@@ -563,6 +566,41 @@ class GpuKernelShap:
 
         phi_full = t.zeros(b, self.n_groups, self.n_out, device=self.device)
 
+        # speculative fast path: when the last call used a captured graph,
+        # assume the varying pattern repeats, replay immediately and verify
+        # the device-computed pattern probe in the SAME sync as the result
+        # (one host sync per call instead of two); mismatch falls through to
+        # the eager path below
+        if (
+            self._graphs_enabled
+            and self._spec is not None
+            and self.linear is not None
+            and self.n_groups <= 62
+        ):
+            sk0, skey = self._spec
+            sb, sns, soff = skey[0], skey[1], skey[3]
+            entry = self._graphs.get((skey[0], skey[1], skey[2]))
+            if entry is not None and sb == b and sns == (nsamples or sns)                     and soff == int(instance_offset):
+                gbool = self._varying_matrix_dev(X_dev)
+                keys = (gbool.long() << t.arange(
+                    self.n_groups, device=self.device)).sum(dim=1)
+                probe = t.stack([(keys == keys[0]).all().long(), keys[0]])
+                graph, x_static, ids_static, gphi, _ = entry
+                if x_static.data_ptr() != X_dev.data_ptr():
+                    x_static.copy_(X_dev)
+                graph.replay()
+                pc = probe.cpu()                    # the single sync
+                timer.mark("spec")
+                if bool(pc[0]) and int(pc[1]) == sk0:
+                    out = gphi.double().cpu().numpy()
+                    timer.mark("d2h")
+                    return [
+                        np.ascontiguousarray(out[:, :, o])
+                        for o in range(self.n_out)
+                    ]
+                # pattern changed: discard the replayed result, run eagerly
+                self._spec = None
+
         # bucket instances by varying-group pattern (benchmark case: 1 bucket)
         if self.n_groups <= 62:
             # fast path: pack each instance's pattern into an int64 key on
@@ -610,6 +648,14 @@ class GpuKernelShap:
                 ):
                     out = self._graph_explain(X_dev, plan0, varying0, instance_offset)
                     if out is not None:
+                        key0 = 0
+                        for g in varying0:
+                            key0 |= 1 << int(g)
+                        self._spec = (
+                            key0,
+                            (b, plan0.nsamples, varying0.tobytes(),
+                             int(instance_offset)),
+                        )
                         timer.mark("graph")
                         return out
 
