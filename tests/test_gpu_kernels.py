@@ -621,3 +621,37 @@ def test_build_diff_and_pack_kernels(ext):
         for si in range(30):
             bits = sum(int(mh[bi, si, k]) << k for k in range(12))
             assert ph[bi, si] == bits
+
+
+def test_speculative_replay_fallback_on_pattern_change():
+    """After the graph path engages (same shape twice), a batch whose varying
+    pattern differs must be detected by the in-sync probe and answered by the
+    eager path — with exact zeros for the fixed groups."""
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor
+
+    rng = np.random.Generator(np.random.Philox(key=[33, 1]))
+    d = 8
+    pred = LinearPredictor.random(d, 2, seed=1)
+    bg = rng.normal(size=(40, d))
+    bg[:, 0] = 0.5                      # constant background column
+    X1 = rng.normal(size=(16, d))       # all groups vary
+    eng = KernelShapEngine(pred, bg, link="logit", seed=0, device="cuda")
+    r1 = eng.shap_values(X1)            # eager (hit 1)
+    r2 = eng.shap_values(X1)            # capture
+    r3 = eng.shap_values(X1)            # speculative replay
+    assert np.allclose(r1[0], r3[0], atol=1e-6)
+    assert eng._gpu._spec is not None
+    # same shape, different pattern: group 0 fixed for every instance
+    X2 = rng.normal(size=(16, d))
+    X2[:, 0] = 0.5
+    r4 = eng.shap_values(X2)
+    assert np.all(r4[0][:, 0] == 0.0) and np.all(r4[1][:, 0] == 0.0)
+    # and local accuracy still holds on the fallback result
+    from distributedkernelshap_amd.core.links import logit
+
+    total = r4[0].sum(axis=1) + eng.expected_value[0]
+    assert np.abs(total - logit(pred(X2))[:, 0]).max() < 1e-3
+    # returning to the original pattern re-engages the graph path
+    r5 = eng.shap_values(X1)
+    assert np.allclose(r5[0], r1[0], atol=1e-6)
