@@ -442,6 +442,10 @@ def test_engine_tracing():
     assert eng.get_trace() is None
 
 
+@pytest.mark.skipif(
+    __import__("os").environ.get("KSHAP_GRAPH") == "0",
+    reason="graph capture disabled via env",
+)
 def test_graph_replay_tracks_new_inputs():
     """hipGraph replay must produce correct phi for NEW instances (the graph
     reads a static input buffer refreshed before each replay)."""
@@ -623,6 +627,10 @@ def test_build_diff_and_pack_kernels(ext):
             assert ph[bi, si] == bits
 
 
+@pytest.mark.skipif(
+    __import__("os").environ.get("KSHAP_GRAPH") == "0",
+    reason="graph capture disabled via env",
+)
 def test_speculative_replay_fallback_on_pattern_change():
     """After the graph path engages (same shape twice), a batch whose varying
     pattern differs must be detected by the in-sync probe and answered by the
