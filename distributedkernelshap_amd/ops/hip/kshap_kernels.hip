@@ -3,25 +3,40 @@
 // Native implementation of the numeric inner loop the reference delegates to
 // shap 0.35.0 (reference call site explainers/kernel_shap.py:250; kernel
 // inventory SURVEY.md §2.4):
-//   K2  fill_random_masks      — counter-based (Philox4x32-10) coalition
-//                                sampling, complement-paired, wave-parallel
-//   K3/K4/K5/K6 fused_predict_linear — MFMA-tiled (v_mfma_f32_16x16x4_f32)
-//                                mask @ diff GEMM with LDS-staged mask /
-//                                background-partial tiles, fused activation
-//                                (none/sigmoid/softmax) and fused weighted
-//                                background reduction -> ey.  The masked-
-//                                background perturbation synthesis is folded
-//                                algebraically: for a linear predictor,
-//                                logits(synth[s,n]) = base[n] + sum_g
-//                                mask[s,g]*(x_part[g]-bg_part[n,g]), so the
-//                                207,200-row synth matrix never touches HBM.
-//   K3' synth_chunk            — explicit masked-background blend for the
-//                                arbitrary-(torch)-predictor path.
-//   K7  wls_solve              — batched constrained weighted-least-squares:
-//                                Gram build over S samples, in-LDS Cholesky,
-//                                back-substitution of the eliminated feature.
+//   K2  fill_random_masks       — counter-based (Philox4x32-10) coalition
+//                                 sampling, complement-paired; 8 waves per
+//                                 instance with ballot/prefix row allocation
+//   K2b pack_masks              — u8 mask rows -> packed u64 bits for the
+//                                 MFMA WLS Gram build
+//   K3/K4/K5/K6 fused_predict_linear / fused_predict_bf16 —
+//                                 MFMA-tiled mask @ diff GEMM
+//                                 (v_mfma_f32_16x16x4_f32, or
+//                                 v_mfma_f32_16x16x32_bf16 with an optional
+//                                 hi+lo split B for fp32-grade results) with
+//                                 LDS-staged diff tiles, the A operand
+//                                 converted in-register from the raw u8
+//                                 masks, fused activation (none / sigmoid /
+//                                 softmax / binary-softmax-from-logit-
+//                                 difference) and fused weighted background
+//                                 reduction -> ey.  The masked-background
+//                                 perturbation synthesis is folded
+//                                 algebraically: for a linear predictor,
+//                                 logits(synth[s,n]) = base[n] + sum_g
+//                                 mask[s,g]*(x_part[g]-bg_part[n,g]), so the
+//                                 207,200-row synth matrix never touches HBM.
+//   build_diff_f32 / build_diff_bf16 — diff images written directly in the
+//                                 fused kernels' operand layouts.
+//   K3' synth_chunk             — explicit masked-background blend for the
+//                                 arbitrary-(torch)-predictor path.
+//   K7  wls_solve_mfma / wls_solve — batched constrained weighted-least-
+//                                 squares: Gram+rhs built on matrix cores
+//                                 (one 16x16x4 MFMA per 4 samples computes
+//                                 both) or scalar path, in-LDS Cholesky,
+//                                 back-substitution of the eliminated
+//                                 feature.
 //
-// All kernels are wave64 / LDS-tiled for CDNA4; fp32 compute throughout.
+// All kernels are wave64 / LDS-tiled for CDNA4; fp32 compute by default,
+// bf16 matrix-core modes opt-in (engine KernelConfig.predict_dtype).
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
