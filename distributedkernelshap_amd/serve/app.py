@@ -190,11 +190,26 @@ def create_app(
 
     @app.api_route("/explain", methods=["GET", "POST"])
     async def explain(request: Request):
+        from fastapi.responses import JSONResponse
+
         t0 = time.perf_counter()
-        payload = await request.json()
+        try:
+            payload = await request.json()
+        except Exception:
+            return JSONResponse({"error": "body must be JSON"}, status_code=400)
+        if not isinstance(payload, dict) or "array" not in payload:
+            return JSONResponse(
+                {"error": "payload must be a JSON object with an 'array' key"},
+                status_code=400,
+            )
         fut: asyncio.Future = asyncio.get_event_loop().create_future()
         await queue.put((payload, fut))
-        body = await fut
+        try:
+            body = await fut
+        except Exception as e:
+            return JSONResponse(
+                {"error": f"explanation failed: {e!r}"}, status_code=400
+            )
         if req_count is not None:
             req_count.inc()
             req_latency.observe(time.perf_counter() - t0)

@@ -113,3 +113,20 @@ def test_metrics_endpoint(served):
         r = client.get("/metrics")
         assert r.status_code == 200
         assert b"kshap_requests_total" in r.content
+
+
+def test_bad_requests_return_400(served):
+    from fastapi.testclient import TestClient
+
+    data, pred, ckw, fkw = served
+    model = BatchKernelShapModel(pred, data.background, ckw, fkw)
+    app = create_app(model, max_batch_size=4, max_wait_ms=5)
+    with TestClient(app) as client:
+        assert client.post("/explain", content=b"not json").status_code == 400
+        assert client.post("/explain", json={"nope": 1}).status_code == 400
+        # wrong feature width -> engine error surfaced as 400, server stays up
+        r = client.post("/explain", json={"array": [[1.0, 2.0]]})
+        assert r.status_code == 400
+        # and a good request still works afterwards
+        ok = client.post("/explain", json={"array": data.X[:1].tolist()})
+        assert ok.status_code == 200
