@@ -168,3 +168,22 @@ def test_linear_fast_path_matches_general(rng):
     svs = slow.shap_values(X)
     for o in range(2):
         assert np.allclose(svf[o], svs[o], atol=1e-10)
+
+
+def test_sigmoid_single_output_predictor(rng):
+    """n_out=1 sigmoid predictor (binary LR parameterised as one logit)."""
+    from distributedkernelshap_amd.core.links import logit
+
+    d = 7
+    w = rng.normal(size=(1, d))
+    pred = LinearPredictor(w, np.zeros(1), activation="sigmoid")
+    bg = rng.normal(size=(25, d))
+    X = rng.normal(size=(3, d))
+    eng = KernelShapEngine(pred, bg, link="logit", seed=0, device="cpu")
+    sv = eng.shap_values(X)
+    assert len(sv) == 1 and sv[0].shape == (3, d)
+    total = sv[0].sum(axis=1) + eng.expected_value[0]
+    assert np.allclose(total, logit(pred(X))[:, 0], atol=1e-9)
+    # for sigmoid+logit link the attribution is the linear closed form
+    expect = (X - bg.mean(axis=0)) * w[0]
+    assert np.allclose(sv[0], expect, atol=1e-6)

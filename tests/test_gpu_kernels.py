@@ -663,3 +663,22 @@ def test_speculative_replay_fallback_on_pattern_change():
     # returning to the original pattern re-engages the graph path
     r5 = eng.shap_values(X1)
     assert np.allclose(r5[0], r1[0], atol=1e-6)
+
+
+def test_engine_gpu_sigmoid_single_output():
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.core.links import logit
+    from distributedkernelshap_amd.models import LinearPredictor
+
+    rng = np.random.Generator(np.random.Philox(key=[41, 2]))
+    d = 7
+    w = rng.normal(size=(1, d))
+    pred = LinearPredictor(w, np.zeros(1), activation="sigmoid")
+    bg = rng.normal(size=(25, d))
+    X = rng.normal(size=(8, d))
+    eng = KernelShapEngine(pred, bg, link="logit", seed=0, device="cuda")
+    sv = eng.shap_values(X)
+    assert len(sv) == 1
+    # sigmoid + logit link on a linear model: linear closed form (fp32 tol)
+    expect = (X - bg.mean(axis=0)) * w[0]
+    assert np.allclose(sv[0], expect, atol=5e-4), np.abs(sv[0] - expect).max()
