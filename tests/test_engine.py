@@ -184,6 +184,11 @@ def test_sigmoid_single_output_predictor(rng):
     assert len(sv) == 1 and sv[0].shape == (3, d)
     total = sv[0].sum(axis=1) + eng.expected_value[0]
     assert np.allclose(total, logit(pred(X))[:, 0], atol=1e-9)
-    # for sigmoid+logit link the attribution is the linear closed form
-    expect = (X - bg.mean(axis=0)) * w[0]
-    assert np.allclose(sv[0], expect, atol=1e-6)
+    # with a SINGLE background row, logit(sigmoid(z)) == z exactly, so the
+    # game is additive and the linear closed form holds (a multi-row
+    # background is NOT linear: logit of the mean of sigmoids)
+    bg1 = rng.normal(size=(1, d))
+    eng1 = KernelShapEngine(pred, bg1, link="logit", seed=0, device="cpu")
+    sv1 = eng1.shap_values(X)
+    expect = (X - bg1[0]) * w[0]
+    assert np.allclose(sv1[0], expect, atol=1e-6)

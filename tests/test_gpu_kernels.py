@@ -674,11 +674,10 @@ def test_engine_gpu_sigmoid_single_output():
     d = 7
     w = rng.normal(size=(1, d))
     pred = LinearPredictor(w, np.zeros(1), activation="sigmoid")
-    bg = rng.normal(size=(25, d))
+    bg = rng.normal(size=(1, d))   # single row: logit(sigmoid(z)) == z
     X = rng.normal(size=(8, d))
     eng = KernelShapEngine(pred, bg, link="logit", seed=0, device="cuda")
     sv = eng.shap_values(X)
     assert len(sv) == 1
-    # sigmoid + logit link on a linear model: linear closed form (fp32 tol)
-    expect = (X - bg.mean(axis=0)) * w[0]
+    expect = (X - bg[0]) * w[0]
     assert np.allclose(sv[0], expect, atol=5e-4), np.abs(sv[0] - expect).max()
