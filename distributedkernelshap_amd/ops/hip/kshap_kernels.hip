@@ -1112,7 +1112,7 @@ void wls_solve_mfma_kernel(
 // chunk round trips.  Same wave-local consumption pattern as the chunked
 // kernel (thread tid stages samples tid, tid+256, ... which its own wave
 // consumes), so no staging barriers are needed.
-#define WLS_FULL_MAX_LDS 47000
+#define WLS_FULL_MAX_LDS 63000
 
 __global__ __launch_bounds__(256)
 void wls_solve_mfma_full_kernel(
