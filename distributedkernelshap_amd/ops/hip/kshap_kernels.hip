@@ -1106,131 +1106,6 @@ void wls_solve_mfma_kernel(
     }
 }
 
-// Full-S variant: when the whole instance's samples fit in LDS
-// (S*(12+4*n_out) bytes), stage them in ONE pass — the per-thread loads
-// issue back-to-back so their latencies overlap, instead of 8 serialized
-// chunk round trips.  Same wave-local consumption pattern as the chunked
-// kernel (thread tid stages samples tid, tid+256, ... which its own wave
-// consumes), so no staging barriers are needed.
-#define WLS_FULL_MAX_LDS 63000
-
-__global__ __launch_bounds__(256)
-void wls_solve_mfma_full_kernel(
-    const uint64_t* __restrict__ packed, // (B, S)
-    const float* __restrict__ kw,        // (B, S)
-    const float* __restrict__ ey_adj,    // (B, S, n_out)
-    const float* __restrict__ total,     // (B, n_out)
-    float* __restrict__ phi,             // (B, M, n_out)
-    int B, int S, int M, int n_out)
-{
-    const int b = blockIdx.x;
-    if (b >= B) return;
-    const int tid = threadIdx.x;
-    const int lane = tid & (WAVE - 1);
-    const int wv = tid >> 6;
-    const int mm = M - 1;
-    const int cols = mm + n_out;
-
-    extern __shared__ float ldsf[];
-    // layout: pk (S u64) | wch (S f32) | ey2 (S*n_out f32) | tails
-    uint64_t* pk = (uint64_t*)ldsf;
-    float* wch = (float*)(pk + S);
-    float* eych = wch + S;
-    float* tails = eych + (size_t)S * n_out;
-    float (*tileA)[16][17] = (float (*)[16][17])tails;  // 4*16*17
-    float* A = tails + 4 * 16 * 17;                   // 16*16
-    float* rhs = A + 16 * 16;                         // 16*WLS_MAX_NOUT
-    float* tot_s = rhs + 16 * WLS_MAX_NOUT;           // WLS_MAX_NOUT
-
-    if (tid < n_out) tot_s[tid] = total[(size_t)b * n_out + tid];
-    __syncthreads();
-
-    const uint64_t* pbase = packed + (size_t)b * S;
-    const float* kwb = kw + (size_t)b * S;
-    const float* eyb = ey_adj + (size_t)b * S * n_out;
-
-    // one staging sweep: thread tid covers samples tid, tid+256, ... — the
-    // exact samples its own wave consumes below (t % 256 -> wave t/64 % 4)
-    for (int i = tid; i < S; i += 256) {
-        uint64_t bits = pbase[i];
-        pk[i] = bits;
-        float w = kwb[i];
-        wch[i] = w;
-        float mlast = (float)((bits >> (M - 1)) & 1ull);
-        for (int o = 0; o < n_out; ++o)
-            eych[(size_t)i * n_out + o] =
-                eyb[(size_t)i * n_out + o] - mlast * tot_s[o];
-    }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-
-    const int arow = lane & 15;
-    const int akk = lane >> 4;
-    f32x4 acc = (f32x4){0, 0, 0, 0};
-    for (int c0 = 0; c0 < S; c0 += WLS_CHUNK) {
-        const int clen = min(WLS_CHUNK, S - c0);
-        const int base = c0 + wv * 64;
-        for (int ks = 0; ks < 64; ks += 4) {
-            int t = base + ks + akk;
-            float a = 0.0f, bv = 0.0f;
-            if (t < c0 + clen) {
-                uint64_t bits = pk[t];
-                float ml = (float)((bits >> (M - 1)) & 1ull);
-                float w = wch[t];
-                if (arow < mm) a = (float)((bits >> arow) & 1ull) - ml;
-                if (arow < mm) bv = w * ((float)((bits >> arow) & 1ull) - ml);
-                else if (arow < cols) bv = w * eych[(size_t)t * n_out + (arow - mm)];
-            }
-            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
-        }
-    }
-#pragma unroll
-    for (int r = 0; r < 4; ++r)
-        tileA[wv][(lane >> 4) * 4 + r][lane & 15] = acc[r];
-    __syncthreads();
-    for (int idx = tid; idx < 16 * 16; idx += 256) {
-        int i = idx / 16, j = idx % 16;
-        float v = tileA[0][i][j] + tileA[1][i][j] + tileA[2][i][j] + tileA[3][i][j];
-        if (j < mm) A[i * mm + j] = v;
-        else if (j < cols && i < mm) rhs[i * n_out + (j - mm)] = v;
-    }
-    __syncthreads();
-    if (tid == 0) {
-        for (int k = 0; k < mm; ++k) {
-            float d = A[k * mm + k];
-            for (int t = 0; t < k; ++t) d -= A[k * mm + t] * A[k * mm + t];
-            d = sqrtf(fmaxf(d, 1e-20f));
-            A[k * mm + k] = d;
-            float inv = 1.0f / d;
-            for (int r = k + 1; r < mm; ++r) {
-                float v = A[r * mm + k];
-                for (int t = 0; t < k; ++t) v -= A[r * mm + t] * A[k * mm + t];
-                A[r * mm + k] = v * inv;
-            }
-        }
-    }
-    __syncthreads();
-    if (tid < n_out) {
-        const int o = tid;
-        float y[16];
-        for (int i = 0; i < mm; ++i) {
-            float v = rhs[i * n_out + o];
-            for (int t = 0; t < i; ++t) v -= A[i * mm + t] * y[t];
-            y[i] = v / A[i * mm + i];
-        }
-        float w[16];
-        float sumw = 0.0f;
-        for (int i = mm - 1; i >= 0; --i) {
-            float v = y[i];
-            for (int t = i + 1; t < mm; ++t) v -= A[t * mm + i] * w[t];
-            w[i] = v / A[i * mm + i];
-        }
-        for (int i = 0; i < mm; ++i) sumw += w[i];
-        float* prow = phi + (size_t)b * M * n_out;
-        for (int i = 0; i < mm; ++i) prow[i * n_out + o] = w[i];
-        prow[(M - 1) * n_out + o] = tot_s[o] - sumw;
-    }
-}
-
 extern "C" int launch_wls_solve(
     const uint8_t* masks, const uint64_t* packed, const float* kw,
     const float* ey_adj, const float* total, float* phi, int B, int S, int M,
@@ -1240,16 +1115,11 @@ extern "C" int launch_wls_solve(
     int mm = M - 1;
     if (mm * (mm + 1) / 2 > 8 * 256) return -1;
     if (mm + n_out <= 16 && packed != nullptr) {
-        size_t full_lds = (size_t)S * (12 + 4 * n_out)
-                          + (4 * 16 * 17 + 16 * 16 + 16 * WLS_MAX_NOUT
-                             + WLS_MAX_NOUT) * 4;
-        if (full_lds <= WLS_FULL_MAX_LDS) {
-            wls_solve_mfma_full_kernel<<<dim3(B), dim3(256), full_lds, stream>>>(
-                packed, kw, ey_adj, total, phi, B, S, M, n_out);
-        } else {
-            wls_solve_mfma_kernel<<<dim3(B), dim3(256), 0, stream>>>(
-                packed, kw, ey_adj, total, phi, B, S, M, n_out);
-        }
+        // (a full-S LDS-staged variant was tried and measured SLOWER —
+        // 222 vs 120 us at S=2072: the 47 KB footprint cut occupancy; the
+        // 8-chunk 17 KB version wins on wave overlap)
+        wls_solve_mfma_kernel<<<dim3(B), dim3(256), 0, stream>>>(
+            packed, kw, ey_adj, total, phi, B, S, M, n_out);
     } else {
         wls_solve_kernel<<<dim3(B), dim3(256), 0, stream>>>(
             masks, kw, ey_adj, total, phi, B, S, M, n_out);
