@@ -128,6 +128,13 @@ def main() -> None:
     )
 
     rank, world = init_distributed()
+    if args.gpus != world:
+        raise SystemExit(
+            f"--gpus {args.gpus} does not match the launched world size "
+            f"{world}: for N>1 launch as `python -m torch.distributed.run "
+            f"--nnodes=1 --nproc-per-node N --master-addr 127.0.0.1 bench.py "
+            f"--gpus N ...` (one rank per GPU over RCCL)"
+        )
     use_cuda = (
         torch.cuda.is_available() if args.device == "auto" else args.device == "cuda"
     )

@@ -68,7 +68,8 @@ def kernel_shap_postprocess_fn(
     ]
 
 
-def _worker_main(worker_id, explainer_type, init_args, init_kwargs, task_q, result_q):
+def _worker_main(worker_id, explainer_type, init_args, init_kwargs, task_q,
+                 result_q, attr_q=None):
     """Worker process: build an explainer replica, serve explain/attr requests.
 
     CUDA replicas are pinned round-robin to the node's GPUs via
@@ -109,10 +110,16 @@ def _worker_main(worker_id, explainer_type, init_args, init_kwargs, task_q, resu
                 )
                 result_q.put(("result", out[0], out[1]))
             elif kind == "attr":
+                # attribute replies go to a dedicated queue so a fetch while
+                # explain results are in flight can never consume (and lose)
+                # a 'result' message
                 _, name = msg
-                result_q.put(("attr", name, replica.return_attribute(name)))
+                (attr_q or result_q).put(
+                    ("attr", name, replica.return_attribute(name))
+                )
         except Exception as e:
-            result_q.put(("error", msg[1] if len(msg) > 1 else None, repr(e)))
+            q = (attr_q or result_q) if kind == "attr" else result_q
+            q.put(("error", msg[1] if len(msg) > 1 else None, repr(e)))
 
 
 class DistributedExplainer:
@@ -147,19 +154,25 @@ class DistributedExplainer:
             "spawn" if explainer_init_kwargs.get("device") == "cuda" else "fork",
         )
         self._ctx = mp.get_context(method)
-        self._task_queues: List[Any] = []
+        self._task_q = None
         self._procs: List[Any] = []
         self._result_q = None
+        self._attr_q = None
         self.create_parallel_pool()
 
     # ------------------------------------------------------------------ #
 
     def create_parallel_pool(self) -> None:
         """Spawn N replica processes (reference :120-128; each ctor ships the
-        pickled predictor + background — the 'broadcast')."""
+        pickled predictor + background — the 'broadcast'). All workers pull
+        from ONE shared task queue: dynamic greedy dispatch, the same load
+        balancing as the reference's ``ActorPool.map_unordered``
+        (``explainers/distributed.py:150-152``) — a slow batch delays only
+        itself, not a statically assigned queue behind it."""
         self._result_q = self._ctx.Queue()
+        self._attr_q = self._ctx.Queue()
+        self._task_q = self._ctx.Queue()
         for wid in range(self.n_workers):
-            tq = self._ctx.Queue()
             p = self._ctx.Process(
                 target=_worker_main,
                 args=(
@@ -167,13 +180,13 @@ class DistributedExplainer:
                     self._explainer_type,
                     self._init_args,
                     self._init_kwargs,
-                    tq,
+                    self._task_q,
                     self._result_q,
+                    self._attr_q,
                 ),
                 daemon=True,
             )
             p.start()
-            self._task_queues.append(tq)
             self._procs.append(p)
         ready = 0
         while ready < self.n_workers:
@@ -193,9 +206,9 @@ class DistributedExplainer:
         cache = self.__dict__.get("_attr_cache", {})
         if item in cache:
             return cache[item]
-        self._task_queues[0].put(("attr", item))
+        self._task_q.put(("attr", item))
         while True:
-            msg = self._result_q.get()
+            msg = self._attr_q.get()
             if msg[0] == "attr" and msg[1] == item:
                 cache[item] = msg[2]
                 return msg[2]
@@ -212,9 +225,7 @@ class DistributedExplainer:
         base = int(kwargs.pop("instance_offset", 0))
         n_batches = len(batches)
         for i, b in enumerate(batches):
-            self._task_queues[i % self.n_workers].put(
-                ("explain", i, b, base + int(offsets[i]), kwargs)
-            )
+            self._task_q.put(("explain", i, b, base + int(offsets[i]), kwargs))
         import queue as _queue
 
         unordered: List[Tuple[int, Any]] = []
@@ -254,16 +265,17 @@ class DistributedExplainer:
     # ------------------------------------------------------------------ #
 
     def shutdown(self) -> None:
-        for q in self._task_queues:
-            try:
-                q.put(None)
-            except Exception:
-                pass
+        if self._task_q is not None:
+            for _ in self._procs:
+                try:
+                    self._task_q.put(None)
+                except Exception:
+                    pass
         for p in self._procs:
             p.join(timeout=5)
             if p.is_alive():
                 p.terminate()
-        self._task_queues = []
+        self._task_q = None
         self._procs = []
 
     def __del__(self):  # best-effort cleanup

@@ -14,8 +14,11 @@ import warnings
 from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple, Union
 
 import numpy as np
+import pandas as pd
+import scipy.sparse as sparse
 
 from ..core.engine import KernelShapEngine
+from ..utils import methdispatch
 from ..interface import (
     DEFAULT_DATA_KERNEL_SHAP,
     DEFAULT_META_KERNEL_SHAP,
@@ -327,42 +330,43 @@ class KernelShap(Explainer, FitMixin):
 
     # ------------------------------------------------------------------ #
 
-    @staticmethod
-    def _coerce_data(data, want_names: bool = True):
+    @methdispatch
+    def _coerce_data(self, data, want_names: bool = True):
         """Normalise supported input containers to (ndarray, names).
 
-        Reference ``_get_data`` singledispatch (``explainers/kernel_shap.py:
-        544-671``) registered shap DenseData, np.ndarray, scipy sparse,
-        DataFrame and Series; here every container collapses to a dense
-        float64 ndarray plus column names when the container carries them.
+        Single-dispatch on the data type via ``utils.methdispatch``, mirroring
+        the reference ``_get_data`` (``explainers/kernel_shap.py:544-671``,
+        dispatch helper ``utils.py:43-64``): registered overloads for
+        DataFrame, Series and scipy sparse below; the base case covers
+        ndarray and any array-like.
         """
-        names = None
-        try:
-            import pandas as pd
-
-            if isinstance(data, pd.Series):
-                names = [str(data.name)] if data.name is not None else None
-                data = data.to_numpy().reshape(1, -1)
-            elif isinstance(data, pd.DataFrame):
-                names = [str(c) for c in data.columns]
-                data = data.to_numpy()
-        except ImportError:  # pragma: no cover
-            pass
-        try:
-            import scipy.sparse as sp
-
-            if sp.issparse(data):
-                logger.warning(
-                    "Sparse background/input densified; KernelSHAP perturbation "
-                    "synthesis operates on dense rows."
-                )
-                data = data.toarray()
-        except ImportError:  # pragma: no cover
-            pass
         arr = np.asarray(data, dtype=np.float64)
         if arr.ndim == 1:
             arr = arr.reshape(1, -1)
+        return (arr, None) if want_names else arr
+
+    @_coerce_data.register(pd.Series)
+    def _coerce_series(self, data, want_names: bool = True):
+        names = [str(data.name)] if data.name is not None else None
+        arr = np.asarray(data.to_numpy(), dtype=np.float64).reshape(1, -1)
         return (arr, names) if want_names else arr
+
+    @_coerce_data.register(pd.DataFrame)
+    def _coerce_frame(self, data, want_names: bool = True):
+        names = [str(c) for c in data.columns]
+        arr = np.asarray(data.to_numpy(), dtype=np.float64)
+        return (arr, names) if want_names else arr
+
+    @_coerce_data.register(sparse.spmatrix)
+    def _coerce_sparse(self, data, want_names: bool = True):
+        logger.warning(
+            "Sparse background/input densified; KernelSHAP perturbation "
+            "synthesis operates on dense rows."
+        )
+        arr = np.asarray(data.toarray(), dtype=np.float64)
+        if arr.ndim == 1:
+            arr = arr.reshape(1, -1)
+        return (arr, None) if want_names else arr
 
     def fit(
         self,
@@ -508,13 +512,21 @@ class KernelShap(Explainer, FitMixin):
                 ]
                 summarised = True
 
-        raw_pred = np.asarray(self.predictor(X))
+        model_out = np.asarray(self.predictor(X))
+        # raw_prediction is exposed on the explainer's link scale — the
+        # reference applies linkfv(self.predictor(X)) (kernel_shap.py:948-950)
+        from ..core.links import convert_to_link
+
+        linkf, _ = convert_to_link(self.link if isinstance(self.link, str) else "identity")
+        raw_pred = linkf(model_out)
         if self.task == "regression":
             prediction = raw_pred.reshape(raw_pred.shape[0], -1)
         elif raw_pred.ndim > 1 and raw_pred.shape[1] > 1:
+            # link functions are monotone: argmax is link-invariant
             prediction = np.argmax(raw_pred, axis=1)
         else:
-            prediction = (raw_pred > 0.5).astype(int).reshape(-1)
+            # single-output classifier: threshold in model (probability) space
+            prediction = (model_out > 0.5).astype(int).reshape(-1)
 
         importances = rank_by_importance(
             shap_values,

@@ -577,10 +577,17 @@ class GpuKernelShap:
             and self.linear is not None
             and self.n_groups <= 62
         ):
-            sk0, skey = self._spec
+            sk0, skey, sm = self._spec
             sb, sns, soff = skey[0], skey[1], skey[3]
             entry = self._graphs.get((skey[0], skey[1], skey[2]))
-            if entry is not None and sb == b and sns == (nsamples or sns)                     and soff == int(instance_offset):
+            # the replay is only valid if THIS call's plan (for the varying
+            # count the probe will verify) resolves to the captured sample
+            # count — nsamples=None must re-resolve through the default, not
+            # wildcard-match the old graph — and its l1 mode stays off
+            spec_plan = self.engine._plan(sm, nsamples) if entry is not None else None
+            if (entry is not None and sb == b and spec_plan.nsamples == sns
+                    and not self._l1_active(spec_plan, l1_reg)
+                    and soff == int(instance_offset)):
                 gbool = self._varying_matrix_dev(X_dev)
                 keys = (gbool.long() << t.arange(
                     self.n_groups, device=self.device)).sum(dim=1)
@@ -655,6 +662,7 @@ class GpuKernelShap:
                             key0,
                             (b, plan0.nsamples, varying0.tobytes(),
                              int(instance_offset)),
+                            m0,
                         )
                         timer.mark("graph")
                         return out

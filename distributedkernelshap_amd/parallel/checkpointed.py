@@ -37,13 +37,17 @@ def explain_checkpointed(
     n_out, g = engine.n_out, engine.n_groups
 
     manifest = {"done": [], "b": b, "n_groups": g, "n_out": n_out,
-                "chunk_instances": chunk_instances, "seed": engine.seed}
+                "chunk_instances": chunk_instances, "seed": engine.seed,
+                "nsamples": nsamples, "l1_reg": repr(l1_reg),
+                "instance_offset": instance_offset}
     if os.path.exists(manifest_path):
         with open(manifest_path) as f:
             old = json.load(f)
-        if (old.get("b") == b and old.get("n_groups") == g
-                and old.get("chunk_instances") == chunk_instances
-                and old.get("seed") == engine.seed):
+        # a restart with different sampling/regularisation parameters must
+        # not mix chunks computed under the old ones
+        if all(old.get(k) == manifest[k]
+               for k in ("b", "n_groups", "chunk_instances", "seed",
+                         "nsamples", "l1_reg", "instance_offset")):
             manifest = old
         # else: incompatible checkpoint -> start over (files rewritten below)
 

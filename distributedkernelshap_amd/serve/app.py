@@ -152,12 +152,32 @@ def create_app(
             _dbg = os.environ.get("KSHAP_TIMING") == "1"
             _t0 = time.perf_counter() if _dbg else 0.0
             try:
-                if isinstance(model, BatchKernelShapModel):
+                if isinstance(model, BatchKernelShapModel) and len(payloads) > 1:
+                    try:
+                        results = await loop.run_in_executor(
+                            None, model.batch, payloads
+                        )
+                    except Exception:
+                        # one malformed co-batched request must not 400 the
+                        # well-formed ones: retry each request individually so
+                        # only the offending ones fail
+                        results = []
+                        for p in payloads:
+                            try:
+                                results.append(
+                                    await loop.run_in_executor(None, model, p)
+                                )
+                            except Exception as e:
+                                results.append(e)
+                elif isinstance(model, BatchKernelShapModel):
                     results = await loop.run_in_executor(None, model.batch, payloads)
                 else:
                     results = []
                     for p in payloads:
-                        results.append(await loop.run_in_executor(None, model, p))
+                        try:
+                            results.append(await loop.run_in_executor(None, model, p))
+                        except Exception as e:
+                            results.append(e)
                 if _dbg:
                     print(
                         f"[kshap-serve] batch={len(batch)} "
@@ -165,7 +185,11 @@ def create_app(
                         flush=True,
                     )
                 for fut, res in zip(futures, results):
-                    if not fut.done():
+                    if fut.done():
+                        continue
+                    if isinstance(res, Exception):
+                        fut.set_exception(res)
+                    else:
                         fut.set_result(res)
             except Exception as e:  # propagate to every waiter
                 for fut in futures:
@@ -200,6 +224,24 @@ def create_app(
         if not isinstance(payload, dict) or "array" not in payload:
             return JSONResponse(
                 {"error": "payload must be a JSON object with an 'array' key"},
+                status_code=400,
+            )
+        # validate shape/dtype BEFORE enqueueing so a malformed request can
+        # never poison the batch it would have been coalesced into
+        try:
+            arr = np.atleast_2d(np.asarray(payload["array"], dtype=np.float64))
+            expected_d = getattr(
+                getattr(model.explainer, "_engine", None), "D", None
+            )
+            if expected_d is not None and arr.shape[1] != expected_d:
+                return JSONResponse(
+                    {"error": f"'array' must have {expected_d} columns, "
+                              f"got {arr.shape[1]}"},
+                    status_code=400,
+                )
+        except (ValueError, TypeError) as e:
+            return JSONResponse(
+                {"error": f"'array' is not a numeric matrix: {e}"},
                 status_code=400,
             )
         fut: asyncio.Future = asyncio.get_event_loop().create_future()

@@ -168,18 +168,39 @@ def test_sample_sharded_equals_sequential(problem):
         assert np.allclose(sv[o], seq[o], rtol=0, atol=1e-8)
 
 
-def test_pool_dead_worker_fails_fast(problem):
-    """A killed worker must raise, not hang (reference quirk SURVEY.md §2.8 /
-    §5.3: map_unordered blocked forever)."""
+def test_pool_dead_worker_tolerated(problem):
+    """With the shared task queue (map_unordered-style dynamic dispatch), a
+    worker that dies while idle is harmless: the survivors drain the queue
+    and the result is still complete and correct."""
     data, pred = problem
     ks = KernelShap(
         pred, link="logit", device="cpu",
         distributed_opts={"n_workers": 2, "batch_size": 2},
     )
     ks.fit(data.background, groups=data.groups)
+    seq = KernelShap(pred, link="logit", device="cpu")
+    seq.fit(data.background, groups=data.groups)
+    expected = seq._explainer._engine.shap_values(data.X)
     # kill one worker behind the pool's back
     ks._explainer._procs[0].terminate()
     ks._explainer._procs[0].join()
+    sv = ks._explainer.get_explanation(data.X)
+    for o in range(2):
+        assert np.allclose(sv[o], expected[o], rtol=0, atol=1e-8)
+
+
+def test_pool_all_workers_dead_fails_fast(problem):
+    """When no worker can make progress the pool must raise, not hang
+    (reference quirk SURVEY.md §2.8 / §5.3: map_unordered blocked forever)."""
+    data, pred = problem
+    ks = KernelShap(
+        pred, link="logit", device="cpu",
+        distributed_opts={"n_workers": 2, "batch_size": 2},
+    )
+    ks.fit(data.background, groups=data.groups)
+    for p in ks._explainer._procs:
+        p.terminate()
+        p.join()
     with pytest.raises(RuntimeError, match="died"):
         ks._explainer.get_explanation(data.X)
 

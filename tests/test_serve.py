@@ -127,6 +127,35 @@ def test_bad_requests_return_400(served):
         # wrong feature width -> engine error surfaced as 400, server stays up
         r = client.post("/explain", json={"array": [[1.0, 2.0]]})
         assert r.status_code == 400
+        # ragged (non-numeric-matrix) array rejected before batching
+        assert client.post(
+            "/explain", json={"array": [[1.0, 2.0], [1.0]]}
+        ).status_code == 400
         # and a good request still works afterwards
         ok = client.post("/explain", json={"array": data.X[:1].tolist()})
         assert ok.status_code == 200
+
+
+def test_malformed_request_does_not_poison_cobatched(served):
+    """A malformed request coalesced with good ones must 400 alone: good
+    requests in the same dynamic batch still return 200 (pre-enqueue
+    validation + per-request fallback in the batcher)."""
+    import concurrent.futures
+
+    from fastapi.testclient import TestClient
+
+    data, pred, ckw, fkw = served
+    model = BatchKernelShapModel(pred, data.background, ckw, fkw)
+    app = create_app(model, max_batch_size=8, max_wait_ms=50)
+    with TestClient(app) as client:
+        def post(payload):
+            return client.post("/explain", json=payload)
+
+        good = {"array": data.X[:1].tolist()}
+        bad = {"array": [[1.0, 2.0]]}          # wrong width
+        with concurrent.futures.ThreadPoolExecutor(4) as ex:
+            futs = [ex.submit(post, good), ex.submit(post, bad),
+                    ex.submit(post, good), ex.submit(post, good)]
+            codes = [f.result().status_code for f in futs]
+        assert codes[1] == 400
+        assert codes[0] == codes[2] == codes[3] == 200
