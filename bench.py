@@ -178,15 +178,20 @@ def main() -> None:
         # instance); the benchmark measures the un-regularised WLS path
         ekw["l1_reg"] = False
 
-    def step() -> np.ndarray:
+    def step():
+        # gather per-instance shap rows to every rank — the reference's
+        # order_result gather, done as ONE device-side fp32
+        # RCCL all_gather_into_tensor over xGMI (no host bounce)
+        if is_distributed():
+            sv = engine.shap_values(
+                X=X_in, instance_offset=rank * args.instances,
+                as_tensor=True, **ekw
+            )
+            counts = [args.instances] * world
+            return allgather_rows(sv, counts)
         sv = engine.shap_values(
             X=X_in, instance_offset=rank * args.instances, **ekw
         )
-        # gather per-instance shap rows (class 0) to every rank — the
-        # reference's order_result gather
-        if is_distributed():
-            counts = [args.instances] * world
-            return allgather_rows(sv[0], counts)
         return sv[0]
 
     def sync():

@@ -172,12 +172,16 @@ class KernelShapEngine:
         l1_reg: Union[str, int, float] = "auto",
         instance_offset: int = 0,
         silent: bool = True,
+        as_tensor: bool = False,
     ) -> List[np.ndarray]:
         """Shapley values for a batch of instances.
 
         Returns a list of ``n_out`` arrays of shape ``(B, n_groups)`` — the
         reference's ``shap_values`` output layout
         (``explainers/kernel_shap.py:250`` / ``build_explanation`` contract).
+        With ``as_tensor=True`` the raw ``(B, n_groups, n_out)`` fp32 torch
+        tensor is returned instead (device-resident on the GPU path) so the
+        distributed result gather never bounces through the host.
 
         ``instance_offset`` keys the per-instance counter RNG so distributed
         shards reproduce the single-process result exactly.
@@ -185,8 +189,13 @@ class KernelShapEngine:
         if self._gpu is not None:
             # torch tensors (incl. pinned/device) pass through untouched
             return self._gpu.shap_values(
-                X, nsamples=nsamples, l1_reg=l1_reg, instance_offset=instance_offset
+                X, nsamples=nsamples, l1_reg=l1_reg,
+                instance_offset=instance_offset, as_tensor=as_tensor,
             )
+        if isinstance(X, (list, tuple)) or type(X).__module__ == "numpy":
+            pass
+        elif hasattr(X, "cpu"):  # torch tensor on the CPU path
+            X = X.cpu().numpy()
         X = _as_2d(X).astype(np.float64)
         b = X.shape[0]
         fx = np.asarray(self.predictor(X))
@@ -199,6 +208,12 @@ class KernelShapEngine:
             phi[i] = self._explain_one(
                 X[i], lfx[i] - lfnull, nsamples, l1_reg, instance_offset + i
             )
+        if as_tensor:
+            # CPU oracle keeps fp64 (bitwise parity with the numpy path);
+            # the GPU engine's as_tensor path is fp32 device-resident
+            import torch
+
+            return torch.from_numpy(phi)
         return [np.ascontiguousarray(phi[:, :, o]) for o in range(self.n_out)]
 
     # ------------------------------------------------------------------ #

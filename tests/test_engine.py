@@ -192,3 +192,21 @@ def test_sigmoid_single_output_predictor(rng):
     sv1 = eng1.shap_values(X)
     expect = (X - bg1[0]) * w[0]
     assert np.allclose(sv1[0], expect, atol=1e-6)
+
+
+def test_shap_values_as_tensor_cpu(rng):
+    """as_tensor=True returns the (B, n_groups, n_out) fp64 torch tensor the
+    distributed gather consumes, matching the per-class list bitwise."""
+    import torch
+
+    d = 6
+    pred = LinearPredictor.random(d, 2, seed=3)
+    bg = rng.normal(size=(15, d))
+    X = rng.normal(size=(4, d))
+    eng = KernelShapEngine(pred, bg, link="logit", seed=0, device="cpu")
+    sv = eng.shap_values(X)
+    t = eng.shap_values(X, as_tensor=True)
+    assert torch.is_tensor(t) and t.dtype == torch.float64
+    assert tuple(t.shape) == (4, d, 2)
+    for o in range(2):
+        assert np.array_equal(t[:, :, o].numpy(), sv[o])
