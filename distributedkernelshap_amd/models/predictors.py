@@ -58,11 +58,12 @@ class LinearPredictor:
         return self.weights, self.bias, self.activation
 
     @classmethod
-    def random(cls, d: int, n_out: int = 2, seed: int = 0, scale: float = 0.5):
+    def random(cls, d: int, n_out: int = 2, seed: int = 0, scale: float = 0.5,
+               activation: str = "softmax"):
         rng = np.random.Generator(np.random.Philox(key=[seed, 0x11EA7]))
         w = rng.normal(0.0, scale, size=(n_out, d))
         b = rng.normal(0.0, 0.1, size=(n_out,))
-        return cls(w, b)
+        return cls(w, b, activation)
 
 
 class TorchPredictor:

@@ -336,6 +336,33 @@ class GpuKernelShap:
         self.ext.fused_predict_bf16(masks, diffB, base, wbg, ey, act)
         return ey
 
+    def _ey_fused_tiled(self, masks, X_dev, varying, vidx_t=None):
+        """Tiled MFMA fused predict for the stress shapes (Mpad>64 or
+        Npad>128): diff streamed through LDS in
+        (k-chunk x 128-col) tiles, per-column-tile partial reductions summed
+        by a deterministic second kernel (no float atomics)."""
+        t = self.torch
+        b, s, m = masks.shape
+        mpad = max(4, (m + 3) // 4 * 4)
+        npad = (self.N + 15) // 16 * 16
+        if vidx_t is None:
+            vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
+        act, oimg = self._act_oimg()
+        diff = self._buf(f"difft{m}", (b, oimg, mpad, npad), zeroed=True)
+        self.ext.build_diff_f32(
+            self._x_part_img(X_dev, act), self._bg_part_img(act), vidx_t, diff
+        )
+        base = t.zeros(oimg, npad, device=self.device)
+        base[:, : self.N] = self._base_img(act)
+        wbg = t.zeros(npad, device=self.device)
+        wbg[: self.N] = self.bg_w
+        n_ntiles = (npad + 127) // 128
+        nacc = 1 if act == 3 else self.n_out
+        partial = self._buf("ftpart", (b, n_ntiles, s, nacc))
+        ey = self._buf("ey", (b, s, self.n_out))
+        self.ext.fused_predict_tiled(masks, diff, base, wbg, partial, ey, act)
+        return ey
+
     def _ey_linear_torch(self, masks, X_dev, varying, s_chunk=4096):
         """Library-GEMM fallback for shapes beyond the fused kernel's limits
         (stress configs: M>64 or N>128). ey = reduce(act(mask @ diff + base))."""
@@ -724,6 +751,9 @@ class GpuKernelShap:
                 elif (kc.fused_predict and mpad <= 64 and npad <= 128
                         and self.n_out in (1, 2, 4)):
                     ey = self._ey_fused_linear(masks, sub_X, varying)
+                elif kc.fused_predict and self.n_out in (1, 2, 4):
+                    # stress shapes stay on the hand-written MFMA path
+                    ey = self._ey_fused_tiled(masks, sub_X, varying)
                 else:
                     ey = self._ey_linear_torch(masks, sub_X, varying)
             else:
@@ -758,6 +788,10 @@ class GpuKernelShap:
                     # withholding the packed masks
                     pk = None if kc.wls_mode == "generic" else packed
                     self.ext.wls_solve(masks, kw, ey_adj, total, phi, pk)
+                elif (kc.wls_mode in ("auto", "mfma") and m <= 513
+                        and self.n_out <= 8):
+                    # stress shapes: MFMA-tiled Gram build + fp64 solve
+                    phi = self._solve_gram(masks, kw, ey_adj, total)
                 else:
                     phi = self._solve_torch(masks, kw, ey_adj, total)
             timer.mark("wls")
@@ -798,6 +832,31 @@ class GpuKernelShap:
         if l1_reg == "auto":
             return frac < 0.2
         return l1_reg not in (None, False, 0)
+
+    def _solve_gram(self, masks, kw, ey_adj, total):
+        """Stress-shape WLS: the S-dependent
+        O(S*(M-1)^2) normal-equation build runs as the hand-written
+        ``wls_gram_kernel`` (16x16 MFMA tiles over packed mask bits, fp64
+        via 256-sample chunked promotion); only the S-independent
+        (M-1)x(M-1) solve uses library fp64. Replaces the fp64 torch bmm of
+        ``_solve_torch`` — which materialised multiple (b, S, M) fp64
+        tensors (~GBs of HBM traffic at the stress config) — with a kernel
+        that reads just the packed bits."""
+        t = self.torch
+        b, s, m = masks.shape
+        w_words = (m + 63) // 64
+        packedw = self._buf(f"packedW{w_words}", (b, s, w_words), t.int64)
+        self.ext.pack_masks_words(masks, packedw)
+        mm = m - 1
+        a64 = self._buf(f"gramA{mm}", (b, mm, mm), t.float64)
+        r64 = self._buf(f"gramR{mm}", (b, mm, self.n_out), t.float64)
+        self.ext.wls_gram(packedw, kw, ey_adj, total, a64, r64)
+        try:
+            w = t.linalg.solve(a64, r64)
+        except Exception:
+            w = t.linalg.lstsq(a64, r64).solution
+        phi_last = total.double()[:, None, :] - w.sum(dim=1, keepdim=True)
+        return t.cat([w, phi_last], dim=1).float()
 
     def _solve_torch(self, masks, kw, ey_adj, total):
         """Batched torch WLS (M > 64 stress configs): normal equations via

@@ -43,6 +43,20 @@ extern "C" int launch_wls_solve(
     const float* ey_adj, const float* total, float* phi, int B, int S, int M,
     int n_out, hipStream_t stream);
 
+extern "C" void launch_pack_masks_words(
+    const uint8_t* masks, uint64_t* packed, int B, int S, int M, int W,
+    hipStream_t stream);
+
+extern "C" int launch_wls_gram(
+    const uint64_t* packed, const float* kw, const float* ey_adj,
+    const float* total, double* A64, double* rhs64, int B, int S, int M,
+    int W, int n_out, hipStream_t stream);
+
+extern "C" int launch_fused_predict_tiled(
+    const uint8_t* masksU, const float* diff, const float* base,
+    const float* wbg, float* partial, float* ey, int B, int S, int M,
+    int Mpad, int Npad, int n_out, int act, hipStream_t stream);
+
 namespace {
 
 #define CHECK_DEV(t) TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be a contiguous device tensor")
@@ -183,6 +197,70 @@ void build_diff_bf16(torch::Tensor xp, torch::Tensor bgp, torch::Tensor vidx,
         m, Npad, split, current_stream());
 }
 
+void pack_masks_words(torch::Tensor masks, torch::Tensor packed) {
+    CHECK_DEV(masks); CHECK_DEV(packed);
+    TORCH_CHECK(masks.dtype() == torch::kUInt8, "masks must be u8");
+    TORCH_CHECK(packed.dim() == 3, "packed must be (B,S,W)");
+    int B = masks.size(0), S = masks.size(1), M = masks.size(2);
+    int W = packed.size(2);
+    TORCH_CHECK(packed.size(0) == B && packed.size(1) == S, "packed shape");
+    TORCH_CHECK(W * 64 >= M, "packed word count vs M");
+    launch_pack_masks_words(
+        masks.data_ptr<uint8_t>(),
+        reinterpret_cast<uint64_t*>(packed.data_ptr<int64_t>()),
+        B, S, M, W, current_stream());
+}
+
+void wls_gram(
+    torch::Tensor packed, torch::Tensor kw, torch::Tensor ey_adj,
+    torch::Tensor total, torch::Tensor A64, torch::Tensor rhs64) {
+    CHECK_DEV(packed); CHECK_DEV(kw); CHECK_DEV(ey_adj); CHECK_DEV(total);
+    CHECK_DEV(A64); CHECK_DEV(rhs64);
+    TORCH_CHECK(packed.dim() == 3, "packed must be (B,S,W)");
+    TORCH_CHECK(A64.dtype() == torch::kFloat64 && rhs64.dtype() == torch::kFloat64);
+    int B = packed.size(0), S = packed.size(1), W = packed.size(2);
+    int mm = A64.size(1), n_out = ey_adj.size(2);
+    TORCH_CHECK(A64.size(0) == B && A64.size(2) == mm, "A64 shape");
+    TORCH_CHECK(rhs64.size(0) == B && rhs64.size(1) == mm
+                && rhs64.size(2) == n_out, "rhs64 shape");
+    TORCH_CHECK(W * 64 >= mm + 1, "packed word count vs M");
+    int rc = launch_wls_gram(
+        reinterpret_cast<const uint64_t*>(packed.data_ptr<int64_t>()),
+        kw.data_ptr<float>(), ey_adj.data_ptr<float>(),
+        total.data_ptr<float>(), A64.data_ptr<double>(),
+        rhs64.data_ptr<double>(), B, S, mm + 1, W, n_out, current_stream());
+    TORCH_CHECK(rc == 0, "wls_gram: unsupported shape (M<=513, n_out<=8)");
+}
+
+void fused_predict_tiled(
+    torch::Tensor masks, torch::Tensor diff, torch::Tensor base,
+    torch::Tensor wbg, torch::Tensor partial, torch::Tensor ey, int64_t act) {
+    CHECK_DEV(masks); CHECK_DEV(diff); CHECK_DEV(base); CHECK_DEV(wbg);
+    CHECK_DEV(partial); CHECK_DEV(ey);
+    TORCH_CHECK(masks.dtype() == torch::kUInt8, "masks must be u8 (B,S,M)");
+    int B = masks.size(0), S = masks.size(1), M = masks.size(2);
+    int Mpad = diff.size(2), Npad = diff.size(3);
+    int n_out = ey.size(2);
+    int oimg = (act == 3) ? 1 : n_out;
+    int nacc = (act == 3) ? 1 : n_out;
+    int n_ntiles = (Npad + 127) / 128;
+    TORCH_CHECK(diff.size(1) == oimg, "diff image count vs act");
+    TORCH_CHECK(M <= Mpad, "M vs Mpad");
+    TORCH_CHECK(ey.size(0) == B && ey.size(1) == S, "ey shape");
+    TORCH_CHECK(base.size(0) == oimg && base.size(1) == Npad, "base shape");
+    TORCH_CHECK(wbg.size(0) == Npad, "wbg shape");
+    TORCH_CHECK(partial.size(0) == B && partial.size(1) == n_ntiles
+                && partial.size(2) == S && partial.size(3) == nacc,
+                "partial shape (B, n_ntiles, S, nacc)");
+    int rc = launch_fused_predict_tiled(
+        masks.data_ptr<uint8_t>(), diff.data_ptr<float>(),
+        base.data_ptr<float>(), wbg.data_ptr<float>(),
+        partial.data_ptr<float>(), ey.data_ptr<float>(), B, S, M, Mpad,
+        Npad, n_out, (int)act, current_stream());
+    TORCH_CHECK(rc == 0, "fused_predict_tiled: unsupported shape "
+                         "(Mpad%4==0, Npad%16==0, n_out in {1,2,4})");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -206,4 +284,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("masks"), pybind11::arg("kw"), pybind11::arg("ey_adj"),
           pybind11::arg("total"), pybind11::arg("phi"),
           pybind11::arg("packed") = pybind11::none());
+    m.def("pack_masks_words", &pack_masks_words,
+          "masks u8 -> (B,S,W) packed u64 words (wide-M Gram build)");
+    m.def("wls_gram", &wls_gram,
+          "tiled MFMA Gram+rhs build (fp64 via chunked promotion) for the "
+          "stress WLS shapes, M up to 513");
+    m.def("fused_predict_tiled", &fused_predict_tiled,
+          "tiled MFMA fused predict for Mpad>64 / Npad>128 (LDS-streamed "
+          "diff chunks, per-column-tile partials + deterministic reduce)");
 }

@@ -1126,3 +1126,475 @@ extern "C" int launch_wls_solve(
     }
     return 0;
 }
+
+// ------------------------------------------------------------------------- //
+// K2b (wide): u8 mask rows -> W-word packed bitfields, M up to 64*W.
+// Feeds the tiled MFMA Gram build for the stress shapes (M up to 513).
+// ------------------------------------------------------------------------- //
+
+__global__ void pack_masks_words_kernel(
+    const uint8_t* __restrict__ masks,  // (B, S, M)
+    uint64_t* __restrict__ packed,      // (B, S, W)
+    size_t n_rows, int M, int W)
+{
+    const size_t row = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (row >= n_rows) return;
+    const uint8_t* src = masks + row * M;
+    for (int w = 0; w < W; ++w) {
+        const int k0 = w * 64;
+        const int k1 = min(M, k0 + 64);
+        uint64_t bits = 0ull;
+        int k = k0;
+        if ((((size_t)(src + k0)) & 3) == 0) {
+            for (; k + 4 <= k1; k += 4) {
+                uint32_t word = *(const uint32_t*)(src + k);
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    bits |= ((uint64_t)((word >> (8 * j)) & 1u)) << (k - k0 + j);
+            }
+        }
+        for (; k < k1; ++k) bits |= ((uint64_t)(src[k] & 1)) << (k - k0);
+        packed[row * W + w] = bits;
+    }
+}
+
+extern "C" void launch_pack_masks_words(
+    const uint8_t* masks, uint64_t* packed, int B, int S, int M, int W,
+    hipStream_t stream)
+{
+    size_t n = (size_t)B * S;
+    pack_masks_words_kernel<<<dim3((unsigned)((n + 255) / 256)), dim3(256), 0,
+                              stream>>>(masks, packed, n, M, W);
+}
+
+// ------------------------------------------------------------------------- //
+// K7 (stress shapes): tiled MFMA Gram + rhs build for mm = M-1 up to 512.
+//
+//   A[i][j]   = sum_s w[s] * e[s][i] * e[s][j]        (i, j < mm)
+//   rhs[i][o] = sum_s w[s] * e[s][i] * ey2[s][o]      (cols j = mm..mm+n_out)
+//   with e[s][i] = mask[s][i] - mask[s][last], ey2 = eyAdj - mask_last*total
+//
+// One workgroup per (instance, 16-row tile strip).  The strip loops 16-col
+// tile blocks (4 at a time); per block it scans S in 256-sample chunks whose
+// packed bits/weights/ey2 are staged in LDS, accumulating each 16x16 tile
+// with v_mfma_f32_16x16x4_f32 and PROMOTING the chunk partial to fp64
+// registers — the in-fp32 accumulation length is <= 256, so the Gram reaches
+// the fp64 torch Cholesky/solve at fp64-grade accuracy despite the Shapley
+// kernel weights spanning orders of magnitude.  Only
+// upper-triangle tile blocks are computed; the symmetric half is mirrored at
+// the write.  The S-dependent O(S*mm^2) work runs on matrix cores; the
+// S-independent (mm x mm) solve stays in library fp64.
+// ------------------------------------------------------------------------- //
+
+#define GRAM_CHUNK 256
+#define GRAM_TJB 4
+#define GRAM_MAX_W 8   // M <= 513
+
+__global__ __launch_bounds__(256)
+void wls_gram_kernel(
+    const uint64_t* __restrict__ packed,  // (B, S, W)
+    const float* __restrict__ kw,         // (B, S)
+    const float* __restrict__ ey_adj,     // (B, S, n_out)
+    const float* __restrict__ total,      // (B, n_out)
+    double* __restrict__ A64,             // (B, mm, mm)
+    double* __restrict__ rhs64,           // (B, mm, n_out)
+    int B, int S, int M, int W, int n_out, int TI)
+{
+    const int b = blockIdx.x / TI;
+    const int ti = blockIdx.x % TI;
+    const int mm = M - 1;
+    const int cols = mm + n_out;
+    const int TJ = (cols + 15) / 16;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wv = tid >> 6;
+    const int arow = lane & 15;
+    const int akk = lane >> 4;
+
+    extern __shared__ char smem[];
+    uint64_t* pk = (uint64_t*)smem;                       // GRAM_CHUNK * W
+    float* wch = (float*)(pk + GRAM_CHUNK * W);           // GRAM_CHUNK
+    float* eych = wch + GRAM_CHUNK;                       // GRAM_CHUNK * n_out
+    double* red = (double*)(eych + GRAM_CHUNK * n_out);   // 4 * 16 * 17
+    __shared__ float tot_s[WLS_MAX_NOUT];
+
+    if (tid < n_out) tot_s[tid] = total[(size_t)b * n_out + tid];
+    __syncthreads();
+
+    const uint64_t* pbase = packed + (size_t)b * S * W;
+    const float* kwb = kw + (size_t)b * S;
+    const float* eyb = ey_adj + (size_t)b * S * n_out;
+
+    const int i = ti * 16 + arow;          // this lane's Gram row (A operand)
+    const int wi = i >> 6, ibit = i & 63;
+    const int lw = (M - 1) >> 6, lb = (M - 1) & 63;  // last-mask bit coords
+
+    for (int tj0 = ti; tj0 < TJ; tj0 += GRAM_TJB) {
+        const int ntj = min(GRAM_TJB, TJ - tj0);
+        double dacc[GRAM_TJB][4];
+        f32x4 acc[GRAM_TJB];
+#pragma unroll
+        for (int tt = 0; tt < GRAM_TJB; ++tt) {
+            acc[tt] = (f32x4){0, 0, 0, 0};
+#pragma unroll
+            for (int r = 0; r < 4; ++r) dacc[tt][r] = 0.0;
+        }
+
+        for (int c0 = 0; c0 < S; c0 += GRAM_CHUNK) {
+            const int clen = min(GRAM_CHUNK, S - c0);
+            __syncthreads();
+            for (int idx = tid; idx < clen * W; idx += 256)
+                pk[idx] = pbase[(size_t)(c0 + idx / W) * W + (idx % W)];
+            for (int idx = tid; idx < clen; idx += 256) {
+                wch[idx] = kwb[c0 + idx];
+                const uint64_t lastw = pbase[(size_t)(c0 + idx) * W + lw];
+                const float ml = (float)((lastw >> lb) & 1ull);
+                for (int o = 0; o < n_out; ++o)
+                    eych[idx * n_out + o] =
+                        eyb[(size_t)(c0 + idx) * n_out + o] - ml * tot_s[o];
+            }
+            __syncthreads();
+            // wave wv covers samples [wv*64, wv*64+64) of the chunk; the
+            // four partial tiles are summed in the cross-wave reduce below
+            const int wlo = wv * 64;
+            const int whi = min(clen, wlo + 64);
+            for (int t4 = wlo; t4 < whi; t4 += 4) {
+                const int t = t4 + akk;
+                const bool tv = t < whi;
+                float a = 0.0f, ml = 0.0f, w = 0.0f;
+                const uint64_t* bits = pk + (size_t)t * W;
+                if (tv) {
+                    ml = (float)((bits[lw] >> lb) & 1ull);
+                    w = wch[t];
+                    if (i < mm) a = (float)((bits[wi] >> ibit) & 1ull) - ml;
+                }
+#pragma unroll
+                for (int tt = 0; tt < GRAM_TJB; ++tt) {
+                    if (tt < ntj) {
+                        const int j = (tj0 + tt) * 16 + arow;
+                        float bv = 0.0f;
+                        if (tv) {
+                            if (j < mm)
+                                bv = w * ((float)((bits[j >> 6] >> (j & 63)) & 1ull) - ml);
+                            else if (j < cols)
+                                bv = w * eych[t * n_out + (j - mm)];
+                        }
+                        acc[tt] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            a, bv, acc[tt], 0, 0, 0);
+                    }
+                }
+            }
+            // promote the chunk partial to fp64 (fp32 run length <= 256)
+#pragma unroll
+            for (int tt = 0; tt < GRAM_TJB; ++tt)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    dacc[tt][r] += (double)acc[tt][r];
+                    acc[tt][r] = 0.0f;
+                }
+        }
+
+        // cross-wave reduce + global write, one 16x16 tile at a time
+        for (int tt = 0; tt < ntj; ++tt) {
+            __syncthreads();
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+                red[((size_t)wv * 16 + (lane >> 4) * 4 + r) * 17 + (lane & 15)] =
+                    dacc[tt][r];
+            __syncthreads();
+            {
+                const int row = tid / 16, col = tid % 16;  // 256 = 16*16
+                double v = red[(0 * 16 + row) * 17 + col]
+                         + red[(1 * 16 + row) * 17 + col]
+                         + red[(2 * 16 + row) * 17 + col]
+                         + red[(3 * 16 + row) * 17 + col];
+                const int gi = ti * 16 + row;
+                const int gj = (tj0 + tt) * 16 + col;
+                if (gi < mm) {
+                    if (gj < mm) {
+                        A64[((size_t)b * mm + gi) * mm + gj] = v;
+                        if (gj != gi && gj > gi)
+                            A64[((size_t)b * mm + gj) * mm + gi] = v;
+                    } else if (gj < cols) {
+                        rhs64[((size_t)b * mm + gi) * n_out + (gj - mm)] = v;
+                    }
+                }
+            }
+        }
+    }
+}
+
+extern "C" int launch_wls_gram(
+    const uint64_t* packed, const float* kw, const float* ey_adj,
+    const float* total, double* A64, double* rhs64, int B, int S, int M,
+    int W, int n_out, hipStream_t stream)
+{
+    if (M < 2 || M > 64 * GRAM_MAX_W + 1 || n_out > WLS_MAX_NOUT || W > GRAM_MAX_W)
+        return -1;
+    const int mm = M - 1;
+    const int TI = (mm + 15) / 16;
+    size_t lds = (size_t)GRAM_CHUNK * W * 8 + GRAM_CHUNK * 4
+               + (size_t)GRAM_CHUNK * n_out * 4 + (size_t)4 * 16 * 17 * 8;
+    wls_gram_kernel<<<dim3(B * TI), dim3(256), lds, stream>>>(
+        packed, kw, ey_adj, total, A64, rhs64, B, S, M, W, n_out, TI);
+    return 0;
+}
+
+// ------------------------------------------------------------------------- //
+// K3-K6 fused, tiled for the stress shapes (Mpad > 64 or Npad > 128): the
+// same mask @ diff MFMA GEMM + activation + weighted background reduction,
+// but the diff image is streamed through LDS in (32 k x 128 n) chunks
+// instead of staged whole, and each workgroup covers a 128-column background
+// tile — its epilogue writes a PARTIAL weighted reduction which a second
+// (deterministic, ordered) kernel sums over column tiles into ey.  Grid is
+// (b, n-tile, s-tile) with s-tile fastest so consecutive workgroups share a
+// diff tile through their XCD's L2.
+// ------------------------------------------------------------------------- //
+
+#define FT_NTILE 8          // 16-col sub-tiles per workgroup = 128 columns
+#define FT_NSTRIDE 144      // 128 + 16: ≡16 mod 32 -> conflict-free k-pairs
+
+template <int NOUT, int ACT>
+__global__ __launch_bounds__(256)
+void fused_predict_tiled_kernel(
+    const uint8_t* __restrict__ masksU, // (B, S, M)
+    const float* __restrict__ diff,     // (B, OIMG, Mpad, Npad)
+    const float* __restrict__ base,     // (OIMG, Npad)
+    const float* __restrict__ wbg,      // (Npad)  0 for padding cols
+    float* __restrict__ partial_out,    // (B, n_ntiles, S, NACC)
+    int B, int S, int M, int Mpad, int Npad)
+{
+    constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
+    constexpr int NACC = (ACT == 3) ? 1 : NOUT;
+    constexpr int KC = (OIMG >= 4) ? 16 : 32;   // LDS diff chunk k-depth
+    const int n_ntiles = (Npad + 127) / 128;
+    const int n_stiles = (S + S_TILE - 1) / S_TILE;
+    const int stile = blockIdx.x % n_stiles;
+    const int nt = (blockIdx.x / n_stiles) % n_ntiles;
+    const int b = blockIdx.x / (n_stiles * n_ntiles);
+    const int n0 = nt * 128;
+    const int ncols = min(128, Npad - n0);
+    const int s0 = stile * S_TILE;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wave = tid >> 6;
+    const int swave = wave * 16;
+    const int arow = lane & 15;
+    const int akcol = lane >> 4;
+
+    extern __shared__ float lds[];
+    float* diff_lds = lds;                          // OIMG * KC * FT_NSTRIDE
+    float* base_lds = diff_lds + OIMG * KC * FT_NSTRIDE;  // OIMG * 128
+    float* wbg_lds = base_lds + OIMG * 128;               // 128
+
+    for (int idx = tid; idx < OIMG * 128; idx += 256) {
+        const int o = idx >> 7, n = idx & 127;
+        base_lds[idx] = (n < ncols) ? base[(size_t)o * Npad + n0 + n] : 0.0f;
+    }
+    for (int idx = tid; idx < 128; idx += 256)
+        wbg_lds[idx] = (idx < ncols) ? wbg[n0 + idx] : 0.0f;
+
+    const float* dsrc = diff + (size_t)b * OIMG * Mpad * Npad;
+    const uint8_t* mlane = masksU + ((size_t)b * S + swave + arow) * M;
+    const float* dlane = diff_lds + akcol * FT_NSTRIDE + arow;
+
+    for (int sub = 0; sub < S_TILE / S_SUB; ++sub) {
+        const int ssub0 = s0 + sub * S_SUB;
+        if (ssub0 >= S) break;
+        const int srow = ssub0 + swave + arow;
+        const bool svalid = srow < S;
+        const uint8_t* mrow = mlane + (size_t)ssub0 * M;
+
+        f32x4 acc[FT_NTILE][OIMG];
+#pragma unroll
+        for (int ct = 0; ct < FT_NTILE; ++ct)
+#pragma unroll
+            for (int o = 0; o < OIMG; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
+
+        for (int kc0 = 0; kc0 < Mpad; kc0 += KC) {
+            __syncthreads();
+            for (int idx = tid; idx < OIMG * KC * 128; idx += 256) {
+                const int n = idx & 127;
+                const int k = (idx >> 7) % KC;
+                const int o = idx / (128 * KC);
+                const int kg = kc0 + k;
+                diff_lds[(o * KC + k) * FT_NSTRIDE + n] =
+                    (kg < Mpad && n < ncols)
+                        ? dsrc[((size_t)o * Mpad + kg) * Npad + n0 + n]
+                        : 0.0f;
+            }
+            __syncthreads();
+            const int kend = min(KC, Mpad - kc0);
+            for (int ks = 0; ks < kend; ks += 4) {
+                const int k = kc0 + ks + akcol;
+                const float a =
+                    (svalid && k < M) ? (float)(mrow[k] & 1) : 0.0f;
+#pragma unroll
+                for (int ct = 0; ct < FT_NTILE; ++ct)
+#pragma unroll
+                    for (int o = 0; o < OIMG; ++o) {
+                        const float bv =
+                            dlane[(o * KC + ks) * FT_NSTRIDE + ct * 16];
+                        acc[ct][o] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            a, bv, acc[ct][o], 0, 0, 0);
+                    }
+            }
+        }
+
+        // epilogue: activation + weighted partial reduction over THIS
+        // column tile (cols beyond ncols carry wbg 0 and contribute nothing)
+        float partialv[NACC][4];
+#pragma unroll
+        for (int o = 0; o < NACC; ++o)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) partialv[o][r] = 0.0f;
+#pragma unroll
+        for (int ct = 0; ct < FT_NTILE; ++ct) {
+            const int n = ct * 16 + arow;
+            const float wn = wbg_lds[n];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float z[OIMG];
+#pragma unroll
+                for (int o = 0; o < OIMG; ++o)
+                    z[o] = acc[ct][o][r] + base_lds[o * 128 + n];
+                float zz[NACC];
+                if (ACT == 3) {
+                    zz[0] = fast_rcp(1.0f + __expf(-z[0]));
+                } else if (ACT == 1) {
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o)
+                        zz[o] = fast_rcp(1.0f + __expf(-z[o]));
+                } else if (ACT == 2 && NOUT == 2) {
+                    const float p1 = fast_rcp(1.0f + __expf(z[0] - z[1]));
+                    zz[0] = 1.0f - p1;
+                    zz[1] = p1;
+                } else if (ACT == 2) {
+                    float mx = z[0];
+#pragma unroll
+                    for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
+                    float sum = 0.0f;
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) {
+                        zz[o] = __expf(z[o] - mx);
+                        sum += zz[o];
+                    }
+                    const float inv = fast_rcp(sum);
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
+                } else {
+#pragma unroll
+                    for (int o = 0; o < NACC; ++o) zz[o] = z[o];
+                }
+#pragma unroll
+                for (int o = 0; o < NACC; ++o) partialv[o][r] += wn * zz[o];
+            }
+        }
+#pragma unroll
+        for (int o = 0; o < NACC; ++o)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float v = partialv[o][r];
+                v += __shfl_xor(v, 1);
+                v += __shfl_xor(v, 2);
+                v += __shfl_xor(v, 4);
+                v += __shfl_xor(v, 8);
+                partialv[o][r] = v;
+            }
+        if (arow == 0) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int ss = ssub0 + swave + akcol * 4 + r;
+                if (ss < S)
+#pragma unroll
+                    for (int o = 0; o < NACC; ++o)
+                        partial_out[(((size_t)b * n_ntiles + nt) * S + ss)
+                                        * NACC + o] = partialv[o][r];
+            }
+        }
+    }
+}
+
+// deterministic column-tile reduction: ey[b,s,:] from the per-tile partials
+// (a fixed summation order — float atomics would break the bitwise-
+// determinism guarantee the GPU tests assert)
+__global__ void reduce_partials_kernel(
+    const float* __restrict__ partial,  // (B, n_ntiles, S, nacc)
+    float* __restrict__ ey,             // (B, S, n_out)
+    size_t total_rows, int S, int n_ntiles, int nacc, int n_out, int act)
+{
+    const size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= total_rows) return;       // total_rows = B * S
+    const size_t b = idx / S;
+    const int s = (int)(idx % S);
+    for (int o = 0; o < nacc; ++o) {
+        float v = 0.0f;
+        for (int nt = 0; nt < n_ntiles; ++nt)
+            v += partial[(((size_t)b * n_ntiles + nt) * S + s) * nacc + o];
+        if (act == 3) {
+            ey[((size_t)b * S + s) * n_out + 0] = 1.0f - v;
+            ey[((size_t)b * S + s) * n_out + n_out - 1] = v;
+        } else {
+            ey[((size_t)b * S + s) * n_out + o] = v;
+        }
+    }
+}
+
+template <int NOUT, int ACT>
+static void launch_ft(
+    const uint8_t* masksU, const float* diff, const float* base,
+    const float* wbg, float* partial, float* ey, int B, int S, int M,
+    int Mpad, int Npad, hipStream_t stream)
+{
+    constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
+    constexpr int NACC = (ACT == 3) ? 1 : NOUT;
+    constexpr int KC = (OIMG >= 4) ? 16 : 32;
+    const int n_ntiles = (Npad + 127) / 128;
+    const int n_stiles = (S + S_TILE - 1) / S_TILE;
+    dim3 grid(B * n_ntiles * n_stiles), block(256);
+    size_t lds = (size_t)(OIMG * KC * FT_NSTRIDE + OIMG * 128 + 128) * 4;
+    fused_predict_tiled_kernel<NOUT, ACT><<<grid, block, lds, stream>>>(
+        masksU, diff, base, wbg, partial, B, S, M, Mpad, Npad);
+    size_t rows = (size_t)B * S;
+    reduce_partials_kernel<<<dim3((unsigned)((rows + 255) / 256)), dim3(256),
+                             0, stream>>>(
+        partial, ey, rows, S, n_ntiles, NACC, NOUT, ACT);
+}
+
+template <int NOUT>
+static void launch_ft_act(
+    const uint8_t* masksU, const float* diff, const float* base,
+    const float* wbg, float* partial, float* ey, int B, int S, int M,
+    int Mpad, int Npad, int act, hipStream_t stream)
+{
+    switch (act) {
+        case 0:
+            launch_ft<NOUT, 0>(masksU, diff, base, wbg, partial, ey, B, S, M, Mpad, Npad, stream);
+            break;
+        case 1:
+            launch_ft<NOUT, 1>(masksU, diff, base, wbg, partial, ey, B, S, M, Mpad, Npad, stream);
+            break;
+        case 3:
+            if constexpr (NOUT == 2)
+                launch_ft<NOUT, 3>(masksU, diff, base, wbg, partial, ey, B, S, M, Mpad, Npad, stream);
+            break;
+        default:
+            launch_ft<NOUT, 2>(masksU, diff, base, wbg, partial, ey, B, S, M, Mpad, Npad, stream);
+            break;
+    }
+}
+
+extern "C" int launch_fused_predict_tiled(
+    const uint8_t* masksU, const float* diff, const float* base,
+    const float* wbg, float* partial, float* ey, int B, int S, int M,
+    int Mpad, int Npad, int n_out, int act, hipStream_t stream)
+{
+    if (Npad % 16 != 0 || Mpad % 4 != 0) return -1;
+    switch (n_out) {
+        case 1: launch_ft_act<1>(masksU, diff, base, wbg, partial, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 2: launch_ft_act<2>(masksU, diff, base, wbg, partial, ey, B, S, M, Mpad, Npad, act, stream); break;
+        case 4: launch_ft_act<4>(masksU, diff, base, wbg, partial, ey, B, S, M, Mpad, Npad, act, stream); break;
+        default: return -1;
+    }
+    return 0;
+}

@@ -323,8 +323,9 @@ def test_engine_gpu_torch_module_path():
 # stress-config paths (library-GEMM fallback + batched torch WLS + module)
 
 def test_engine_gpu_stress_paths():
-    """M=200 > 64 forces _ey_linear_torch + _solve_torch; local accuracy and
-    CPU-oracle agreement on a reduced stress shape."""
+    """M=200 > 64 dispatches to the tiled stress kernels
+    (fused_predict_tiled + wls_gram); local accuracy and CPU-oracle
+    agreement on a reduced stress shape."""
     from distributedkernelshap_amd.core.engine import KernelShapEngine
     from distributedkernelshap_amd.core.links import logit
     from distributedkernelshap_amd.models import LinearPredictor, make_tabular
@@ -340,6 +341,79 @@ def test_engine_gpu_stress_paths():
     for o in range(2):
         total = sv[o].sum(axis=1) + eng.expected_value[o]
         assert np.abs(total - fx[:, o]).max() < 2e-3
+    # CPU fp64 oracle agreement (same masks via the counter RNG)
+    cpu = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cpu",
+    )
+    sv_c = cpu.shap_values(data.X, nsamples=2048, l1_reg=False)
+    for o in range(2):
+        err = np.abs(sv[o] - sv_c[o]).max()
+        assert err < 2e-3, err
+
+
+def test_wls_gram_vs_fp64_reference(ext):
+    """Tiled MFMA Gram+rhs build (stress WLS): matches the fp64 torch normal
+    equations to chunked-promotion accuracy, across the M=257 word boundary
+    and with kernel weights spanning orders of magnitude."""
+    g = torch.Generator(device="cuda").manual_seed(17)
+    for (b, s, m, n_out) in [(3, 2048, 100, 2), (2, 4096, 257, 2),
+                             (2, 1000, 70, 1)]:
+        masks = (torch.rand(b, s, m, generator=g, device="cuda") > 0.5).to(
+            torch.uint8
+        )
+        # Shapley-like dynamic range
+        kw = 10.0 ** (
+            -4.0 * torch.rand(b, s, generator=g, device="cuda")
+        )
+        ey = torch.randn(b, s, n_out, generator=g, device="cuda")
+        total = torch.randn(b, n_out, generator=g, device="cuda")
+        w_words = (m + 63) // 64
+        packed = torch.empty(b, s, w_words, dtype=torch.int64, device="cuda")
+        ext.pack_masks_words(masks, packed)
+        mm = m - 1
+        a64 = torch.empty(b, mm, mm, dtype=torch.float64, device="cuda")
+        r64 = torch.empty(b, mm, n_out, dtype=torch.float64, device="cuda")
+        ext.wls_gram(packed, kw, ey, total, a64, r64)
+        # fp64 reference
+        z = masks.double()
+        last = z[:, :, -1:]
+        etmp = z[:, :, :-1] - last
+        ey2 = ey.double() - last * total.double()[:, None, :]
+        wz = etmp * kw.double()[:, :, None]
+        a_ref = torch.bmm(wz.transpose(1, 2), etmp)
+        r_ref = torch.bmm(wz.transpose(1, 2), ey2)
+        ea = (a64 - a_ref).abs().max().item()
+        er = (r64 - r_ref).abs().max().item()
+        scale = a_ref.abs().max().item() + 1.0
+        assert ea < 1e-4 * scale, (m, ea, scale)
+        assert er < 1e-4 * scale, (m, er)
+
+
+def test_fused_predict_tiled_vs_torch():
+    """Tiled fused predict (M>64 / N>128) matches the library-GEMM fallback
+    on the same device masks, for sigmoid, binary softmax and identity."""
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor, make_tabular
+
+    for act, n_out, m, n_bg in [("softmax", 2, 70, 200), ("sigmoid", 2, 150, 300),
+                                ("none", 1, 80, 130)]:
+        data = make_tabular(n_features=m, n_instances=3, n_background=n_bg,
+                            seed=3)
+        pred = LinearPredictor.random(m, n_out, seed=3, activation=act)
+        eng = KernelShapEngine(
+            pred, data.background, groups=data.groups, link="identity",
+            seed=0, device="cuda",
+        )
+        gpu = eng._gpu
+        varying = np.arange(m)
+        plan = eng._plan(m, 1024)
+        masks, _kw = gpu._device_masks(plan, np.arange(3))
+        X_dev = torch.tensor(data.X, dtype=torch.float32, device="cuda")
+        ey_tiled = gpu._ey_fused_tiled(masks, X_dev, varying).clone()
+        ey_ref = gpu._ey_linear_torch(masks, X_dev, varying)
+        err = (ey_tiled - ey_ref).abs().max().item()
+        assert err < 5e-5, (act, n_out, m, n_bg, err)
 
 
 def test_engine_gpu_mlp_module_path_local_accuracy():
