@@ -104,10 +104,17 @@ def main() -> None:
     p.add_argument("--background", type=int, default=None)
     p.add_argument("--nsamples", type=int, default=None)
     p.add_argument("--device", default="auto", choices=["auto", "cuda", "cpu"])
-    p.add_argument("--dtype", default="fp32", choices=["fp32", "bf16x2", "bf16"],
+    p.add_argument("--dtype", default="fp32",
+                   choices=["fp32", "bf16x2", "bf16", "fp64"],
                    help="GPU predict compute mode: fp32 MFMA (default), "
                         "bf16 matrix cores with hi+lo split (fp32-grade), "
-                        "or plain bf16 (fastest)")
+                        "plain bf16 (fastest), or fp64 (full-double "
+                        "verification mode, linear predictors)")
+    p.add_argument("--selfcheck", type=int, default=64, metavar="K",
+                   help="after timing, re-explain the first K instances in "
+                        "the fp64 device mode (same masks) and report "
+                        "max_phi_err_vs_fp64 in the JSON; 0 disables "
+                        "(linear-predictor configs only)")
     args = p.parse_args()
     defaults = CONFIG_DEFAULTS[args.config]
     if args.instances is None:
@@ -222,6 +229,34 @@ def main() -> None:
     n_gpus = world
     ms_per_step = elapsed / args.steps * 1000.0
     value = n_gpus * args.instances * args.steps / elapsed
+
+    # precision self-certification: re-explain a prefix
+    # of this rank's shard in the fp64 device mode — identical masks via the
+    # counter RNG — and report the measured fp32-pipeline error bound
+    max_phi_err = None
+    if (args.selfcheck and use_cuda and args.dtype != "fp64"
+            and isinstance(pred, LinearPredictor)):
+        k = min(args.selfcheck, args.instances)
+        off = rank * args.instances
+        sv_fast = engine.shap_values(X[:k], instance_offset=off, **ekw)
+        eng64 = KernelShapEngine(
+            pred, background, groups=groups, link="logit", seed=0,
+            device=device, kernels=KernelConfig(predict_dtype="fp64"),
+        )
+        sv_64 = eng64.shap_values(X[:k], instance_offset=off, **ekw)
+        max_phi_err = float(
+            max(np.abs(sv_fast[o] - sv_64[o]).max() for o in range(len(sv_64)))
+        )
+        if is_distributed():
+            import torch.distributed as dist
+
+            e = torch.tensor(
+                [max_phi_err], dtype=torch.float64,
+                device="cuda" if dist.get_backend() == "nccl" else "cpu",
+            )
+            dist.all_reduce(e, op=dist.ReduceOp.MAX)
+            max_phi_err = float(e.item())
+
     if rank == 0:
         metric = (
             "explanations/sec (2560 inst, 100-sample background)"
@@ -242,7 +277,9 @@ def main() -> None:
                             if args.config == "adult" else None),
             "dtype": ("fp32" if args.dtype == "fp32" or device == "cpu"
                       else ("bf16x2 (hi+lo split, fp32-grade)"
-                            if args.dtype == "bf16x2" else "bf16")),
+                            if args.dtype == "bf16x2"
+                            else args.dtype)),
+            "max_phi_err_vs_fp64": max_phi_err,
             "data": "synthetic",
             "config": {
                 "model": model_desc,

@@ -72,7 +72,9 @@ class KernelConfig:
     # fp32: f32 MFMA (default).  bf16x2: bf16 matrix cores with a hi+lo
     # split B operand (error ~2^-16, fp32-grade, ~6x less MFMA issue time).
     # bf16: single bf16 image (fastest, ~0.4% relative ey error).
-    predict_dtype: str = "fp32"  # fp32 | bf16x2 | bf16
+    # fp64: full-double verification mode (linear predictors) — the
+    # reference's numpy-fp64 arithmetic, used by bench.py's self-check.
+    predict_dtype: str = "fp32"  # fp32 | bf16x2 | bf16 | fp64
     synth_chunk_rows: int = 1 << 19
 
 

@@ -363,6 +363,49 @@ class GpuKernelShap:
         self.ext.fused_predict_tiled(masks, diff, base, wbg, partial, ey, act)
         return ey
 
+    def _ey_linear_f64(self, masks, X_dev, varying, s_chunk=2048):
+        """fp64 verification path (predict_dtype='fp64'):
+        the SAME device coalition masks evaluated end-to-end in fp64 (weights,
+        diff, GEMM, activation, background reduction all double). Perf is
+        irrelevant — this mode exists so bench.py can print a measured
+        ``max_phi_err_vs_fp64`` for the fp32 pipeline, making every headline
+        number self-certifying (the reference computes in numpy fp64
+        throughout, shap 0.35.0 via ``explainers/kernel_shap.py:250``)."""
+        t = self.torch
+        b, s, m = masks.shape
+        W = self.linear["W"].double()
+        bias = self.linear["b"].double()
+        bg = self.bg.double()
+        X64 = X_dev.double()
+        contrib_x = X64[:, :, None] * W.T[None, :, :]
+        x_part = t.zeros(b, self.n_groups, self.n_out, dtype=t.float64,
+                         device=self.device)
+        x_part.index_add_(1, self.col_group, contrib_x)
+        contrib_b = bg[:, :, None] * W.T[None, :, :]
+        bg_part = t.zeros(self.N, self.n_groups, self.n_out, dtype=t.float64,
+                          device=self.device)
+        bg_part.index_add_(1, self.col_group, contrib_b)
+        vidx = t.tensor(varying, dtype=t.int64, device=self.device)
+        diff = (x_part[:, vidx].permute(0, 2, 1)[:, :, :, None]
+                - bg_part[:, vidx].permute(2, 1, 0)[None])       # (b,o,m,N)
+        diff = diff.permute(0, 2, 3, 1).reshape(b, m, self.N * self.n_out)
+        base = bg @ W.T + bias                                    # (N, o)
+        wbg64 = self.bg_w.double()
+        ey = t.empty(b, s, self.n_out, dtype=t.float64, device=self.device)
+        for lo in range(0, s, s_chunk):
+            hi = min(lo + s_chunk, s)
+            logits = t.bmm(masks[:, lo:hi].double(), diff)
+            logits = logits.view(b, hi - lo, self.N, self.n_out) + base[None, None]
+            a = self.linear["act"]
+            if a == 1:
+                p = t.sigmoid(logits)
+            elif a == 2:
+                p = t.softmax(logits, dim=-1)
+            else:
+                p = logits
+            ey[:, lo:hi] = t.einsum("bsno,n->bso", p, wbg64)
+        return ey
+
     def _ey_linear_torch(self, masks, X_dev, varying, s_chunk=4096):
         """Library-GEMM fallback for shapes beyond the fused kernel's limits
         (stress configs: M>64 or N>128). ey = reduce(act(mask @ diff + base))."""
@@ -604,13 +647,17 @@ class GpuKernelShap:
             X_dev = t.tensor(X, dtype=t.float32, device=self.device)
         timer.mark("h2d")
 
+        fp64 = self.engine.kernels.predict_dtype == "fp64"
         fx = self._predict_rows_f64(X_dev)              # (B, n_out) fp64
         lfx = self._link(fx)
         lfnull64 = self._link(self.fnull.double())
-        total_all = (lfx - lfnull64[None, :]).float()   # (B, n_out)
+        total_all = (lfx - lfnull64[None, :])           # (B, n_out) fp64
+        if not fp64:
+            total_all = total_all.float()
         lfnull = lfnull64.float()
 
-        phi_full = t.zeros(b, self.n_groups, self.n_out, device=self.device)
+        phi_full = t.zeros(b, self.n_groups, self.n_out, device=self.device,
+                           dtype=t.float64 if fp64 else t.float32)
 
         # speculative fast path: when the last call used a captured graph,
         # assume the varying pattern repeats, replay immediately and verify
@@ -619,6 +666,7 @@ class GpuKernelShap:
         # the eager path below
         if (
             self._graphs_enabled
+            and not fp64
             and self._spec is not None
             and self.linear is not None
             and self.n_groups <= 62
@@ -687,7 +735,8 @@ class GpuKernelShap:
         timer.mark("bucket")
 
         kc = self.engine.kernels
-        if self._graphs_enabled and uniq.shape[0] == 1 and self.linear is not None \
+        if self._graphs_enabled and not fp64 and uniq.shape[0] == 1 \
+                and self.linear is not None \
                 and kc.fused_predict and kc.wls_mode in ("auto", "mfma"):
             varying0 = np.nonzero(uniq[0])[0]
             m0 = len(varying0)
@@ -733,6 +782,15 @@ class GpuKernelShap:
             masks, kw = self._device_masks(plan, gids)
             timer.mark("masks")
             sub_X = X_dev[ids_t]
+            if fp64:
+                phi = self._bucket_fp64(
+                    plan, masks, sub_X, varying, ids_t, total_all, lfnull64,
+                    l1_reg,
+                )
+                timer.mark("wls")
+                vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
+                phi_full[ids_t[:, None], vidx_t[None, :]] = phi
+                continue
             packed = None
             if m <= 64:
                 # packed u64 masks feed the MFMA WLS Gram build
@@ -833,6 +891,38 @@ class GpuKernelShap:
             return frac < 0.2
         return l1_reg not in (None, False, 0)
 
+    def _bucket_fp64(self, plan, masks, sub_X, varying, ids_t, total_all,
+                     lfnull64, l1_reg):
+        """One bucket of the fp64 verification pipeline: fp64 ey on the same
+        device masks, exact fp64 kernel weights (rebuilt from the plan's
+        fp64 arrays, not the fp32 device copy), fp64 link and WLS."""
+        t = self.torch
+        if self.linear is None:
+            raise TypeError(
+                "predict_dtype='fp64' is the linear-predictor verification "
+                "mode; torch-module predictors have no fp64 reference "
+                "(convert the module to double and use device='cpu' instead)."
+            )
+        b = sub_X.shape[0]
+        ey = self._ey_linear_f64(masks, sub_X, varying)
+        kw64 = t.from_numpy(
+            np.concatenate([
+                plan.enum_weights,
+                np.full(plan.n_random,
+                        plan.weight_left / max(plan.n_random, 1)),
+            ])
+        ).to(self.device)
+        kw = kw64[None, :].expand(b, plan.nsamples).contiguous()
+        if self.link_name == "identity":
+            ey_adj = ey - lfnull64[None, None, :]
+        else:
+            p = ey.clamp(1e-15, 1.0 - 1e-15)
+            ey_adj = t.log(p / (1.0 - p)) - lfnull64[None, None, :]
+        total = total_all[ids_t].contiguous()
+        if self._l1_active(plan, l1_reg):
+            return self._solve_host_l1(masks, kw, ey_adj, total, l1_reg).double()
+        return self._solve_torch(masks, kw, ey_adj, total, out_dtype=t.float64)
+
     def _solve_gram(self, masks, kw, ey_adj, total):
         """Stress-shape WLS: the S-dependent
         O(S*(M-1)^2) normal-equation build runs as the hand-written
@@ -858,12 +948,12 @@ class GpuKernelShap:
         phi_last = total.double()[:, None, :] - w.sum(dim=1, keepdim=True)
         return t.cat([w, phi_last], dim=1).float()
 
-    def _solve_torch(self, masks, kw, ey_adj, total):
-        """Batched torch WLS (M > 64 stress configs): normal equations via
-        bmm + torch.linalg.solve, in fp64 — the Shapley kernel weights span
+    def _solve_torch(self, masks, kw, ey_adj, total, out_dtype=None):
+        """Batched torch WLS fallback: normal equations via bmm +
+        torch.linalg.solve, in fp64 — the Shapley kernel weights span
         several orders of magnitude, so the Gram matrix at M~200 is too
         ill-conditioned for fp32 normal equations (MI355X fp64 is cheap
-        relative to this cold path)."""
+        relative to this cold path). Also the fp64 verification solver."""
         t = self.torch
         z = masks.double()
         last = z[:, :, -1:]
@@ -874,7 +964,7 @@ class GpuKernelShap:
         r = t.bmm(wz.transpose(1, 2), ey2)               # (b, m-1, o)
         w = t.linalg.solve(a, r)
         phi_last = total.double()[:, None, :] - w.sum(dim=1, keepdim=True)
-        return t.cat([w, phi_last], dim=1).float()
+        return t.cat([w, phi_last], dim=1).to(out_dtype or t.float32)
 
     def _solve_host_l1(self, masks, kw, ey_adj, total, l1_reg):
         """Cold path: l1 feature selection + solve on host, per instance."""
