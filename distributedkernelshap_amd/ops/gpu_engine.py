@@ -828,7 +828,11 @@ class GpuKernelShap:
                 ey_adj = ey.sub_(lfnull[None, None, :])
             total = total_all[ids_t].contiguous()
             if self._l1_active(plan, l1_reg):
-                phi = self._solve_host_l1(masks, kw, ey_adj, total, l1_reg)
+                if kc.l1_device:
+                    support = self._l1_select_batched(masks, kw, ey_adj, l1_reg)
+                    phi = self._solve_selected(masks, kw, ey_adj, total, support)
+                else:
+                    phi = self._solve_host_l1(masks, kw, ey_adj, total, l1_reg)
             else:
                 phi = self._buf("phi", (len(ids), m, self.n_out))
                 # fp32 Gram conditioning degrades with M (Shapley kernel
@@ -965,6 +969,121 @@ class GpuKernelShap:
         w = t.linalg.solve(a, r)
         phi_last = total.double()[:, None, :] - w.sum(dim=1, keepdim=True)
         return t.cat([w, phi_last], dim=1).to(out_dtype or t.float32)
+
+    def _l1_select_batched(self, masks, kw, ey_adj, l1_reg):
+        """K8 on device: whole-batch L1 pre-selection.
+
+        The weighted normal equations (Gram + correlations) are built by the
+        MFMA ``wls_gram`` kernel called with a phantom last feature (bit m is
+        always 0, so the constraint elimination is a no-op and the kernel
+        returns the RAW m x m Gram); the LARS/lasso path then runs as ONE
+        batched torch fp64 iteration over all instances
+        (``core.lars.batched_lars_select``, selection-identical to the
+        sklearn calls of the CPU oracle) — no per-instance host loop.
+        Returns a (b, m) bool support tensor."""
+        from ..core.lars import batched_lars_select
+
+        t = self.torch
+        b, s, m = masks.shape
+        if l1_reg == "auto":
+            if not getattr(self, "_warned_l1_auto", False):
+                logger.warning(
+                    "l1_reg='auto' engaged on a GPU batch (sampled fraction "
+                    "of the 2^%d coalitions < 0.2): running batched device "
+                    "LARS (AIC) pre-selection over all %d instances. For "
+                    "dense problems pass l1_reg=False to skip selection "
+                    "entirely.", m, b,
+                )
+                self._warned_l1_auto = True
+            mode, nf, alpha = "aic", None, None
+        elif l1_reg in ("aic", "bic"):
+            mode, nf, alpha = l1_reg, None, None
+        elif isinstance(l1_reg, str) and l1_reg.startswith("num_features("):
+            mode, nf, alpha = ("num_features",
+                               int(l1_reg[len("num_features("):-1]), None)
+        elif isinstance(l1_reg, (int, float)) and not isinstance(l1_reg, bool):
+            mode, nf, alpha = "alpha", None, float(l1_reg)
+        else:
+            raise ValueError(f"Unsupported l1_reg: {l1_reg!r}")
+
+        w_words = (m + 64) // 64          # phantom last bit m -> W*64 >= m+1
+        packedw = self._buf(f"packedL{w_words}", (b, s, w_words), t.int64)
+        self.ext.pack_masks_words(masks, packedw)
+        g64 = self._buf(f"l1G{m}", (b, m, m), t.float64)
+        r64 = self._buf(f"l1R{m}", (b, m, self.n_out), t.float64)
+        zero_tot = self._buf("l1tot", (b, self.n_out), zeroed=True)
+        self.ext.wls_gram(packedw, kw, ey_adj, zero_tot, g64, r64)
+        c64 = r64[:, :, 0].contiguous()
+        y = ey_adj[:, :, 0]
+        yty = (kw.double() * y.double() * y.double()).sum(1)
+        zbar = ybar = None
+        if mode != "num_features":
+            # sklearn Lasso/LassoLarsIC centre the (weighted) design
+            sqw = kw.clamp_min(0).sqrt()
+            zsum = t.zeros(b, m, dtype=t.float64, device=self.device)
+            for lo in range(0, s, 8192):
+                hi = min(lo + 8192, s)
+                zsum += t.bmm(
+                    sqw[:, lo:hi].unsqueeze(1).double(),
+                    masks[:, lo:hi].double(),
+                ).squeeze(1)
+            zbar = zsum / s
+            ybar = (sqw.double() * y.double()).sum(1) / s
+        return batched_lars_select(
+            g64, c64, yty, n_samples=s, mode=mode, num_features=nf,
+            alpha=alpha, zbar=zbar, ybar=ybar,
+        )
+
+    def _solve_selected(self, masks, kw, ey_adj, total, support):
+        """Constrained WLS over per-instance selected supports: instances are
+        grouped by identical support (the common case) and each group solved
+        with the normal kernel dispatch on the column-sliced masks."""
+        t = self.torch
+        b, s, m = masks.shape
+        phi = t.zeros(b, m, self.n_out, device=self.device)
+        sup_np = support.cpu().numpy().astype(bool)
+        groups: dict = {}
+        for i in range(b):
+            groups.setdefault(sup_np[i].tobytes(), []).append(i)
+        for idxs in groups.values():
+            sel = np.nonzero(sup_np[idxs[0]])[0]
+            ids_t = t.tensor(idxs, dtype=t.int64, device=self.device)
+            tot_g = total[ids_t].contiguous()
+            if len(sel) == 0:
+                # empty selection: shap solves over all features
+                phi[ids_t] = self._solve_for(
+                    masks[ids_t].contiguous(), kw[ids_t].contiguous(),
+                    ey_adj[ids_t].contiguous(), tot_g,
+                )
+                continue
+            if len(sel) == 1:
+                phi[ids_t, int(sel[0])] = tot_g
+                continue
+            sel_t = t.tensor(sel, dtype=t.int64, device=self.device)
+            sub = masks[ids_t][:, :, sel_t].contiguous()
+            phi_sub = self._solve_for(
+                sub, kw[ids_t].contiguous(), ey_adj[ids_t].contiguous(), tot_g
+            )
+            phi[ids_t[:, None], sel_t[None, :]] = phi_sub
+        return phi
+
+    def _solve_for(self, masks, kw, ey_adj, total):
+        """WLS solve dispatch for arbitrary (sub-)shapes."""
+        t = self.torch
+        b, s, m = masks.shape
+        kc = self.engine.kernels
+        if (kc.wls_mode != "torch" and 2 <= m <= 64 and self.n_out <= 8
+                and (m - 1 + self.n_out <= 16 or m <= 24)):
+            phi = t.empty(b, m, self.n_out, device=self.device)
+            packed = None
+            if m - 1 + self.n_out <= 16 and kc.wls_mode != "generic":
+                packed = t.empty(b, s, dtype=t.int64, device=self.device)
+                self.ext.pack_masks(masks, packed)
+            self.ext.wls_solve(masks, kw, ey_adj, total, phi, packed)
+            return phi
+        if kc.wls_mode in ("auto", "mfma") and m <= 513 and self.n_out <= 8:
+            return self._solve_gram(masks, kw, ey_adj, total)
+        return self._solve_torch(masks, kw, ey_adj, total)
 
     def _solve_host_l1(self, masks, kw, ey_adj, total, l1_reg):
         """Cold path: l1 feature selection + solve on host, per instance."""

@@ -587,7 +587,7 @@ def test_engine_gpu_mixed_varying_buckets():
 
 
 def test_engine_gpu_l1_path():
-    """Explicit l1_reg routes through the host LARS path on GPU inputs."""
+    """Explicit l1_reg routes through the batched device LARS on GPU."""
     from distributedkernelshap_amd.core.engine import KernelShapEngine
     from distributedkernelshap_amd.models import LinearPredictor, make_adult_like
 
@@ -600,6 +600,36 @@ def test_engine_gpu_l1_path():
     sv = eng.shap_values(data.X, l1_reg="num_features(5)")
     assert sv[0].shape == (4, 12)
     assert (np.abs(sv[0]) > 1e-12).sum(axis=1).max() <= 6
+
+
+def test_engine_gpu_l1_batched_matches_host_sklearn():
+    """The batched device LARS (l1_device=True) selects the same features —
+    and therefore the same phi — as the per-instance host sklearn path, for
+    aic, num_features and float-alpha modes."""
+    from distributedkernelshap_amd.config import KernelConfig
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor, make_adult_like
+
+    data = make_adult_like(n_instances=5, n_background=30, seed=8)
+    pred = LinearPredictor.random(data.X.shape[1], 2, seed=8)
+    dev_eng = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cuda",
+    )
+    host_eng = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cuda", kernels=KernelConfig(l1_device=False),
+    )
+    for reg in ["aic", "num_features(4)", 0.01]:
+        sv_d = dev_eng.shap_values(data.X, l1_reg=reg)
+        sv_h = host_eng.shap_values(data.X, l1_reg=reg)
+        for o in range(2):
+            nz_d = np.abs(sv_d[o]) > 1e-12
+            nz_h = np.abs(sv_h[o]) > 1e-12
+            assert np.array_equal(nz_d, nz_h), (reg, o)
+            assert np.allclose(sv_d[o], sv_h[o], atol=5e-3), (
+                reg, o, np.abs(sv_d[o] - sv_h[o]).max()
+            )
 
 
 def test_gpu_pipeline_bitwise_deterministic():
