@@ -80,6 +80,11 @@ class KernelConfig:
     # LARS (core.lars), False = per-instance host sklearn (the CPU oracle's
     # path, kept as a cross-check)
     l1_device: bool = True
+    # torch-module predictor path (mlp/resnet configs): bf16 autocast around
+    # the module forward ("bf16") and channels-last weight layout for conv
+    # nets — the predict-bound configs' main levers on MI355X matrix cores
+    module_autocast: str = "off"  # off | bf16
+    module_channels_last: bool = True
 
 
 @dataclass

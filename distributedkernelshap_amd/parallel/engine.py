@@ -114,18 +114,22 @@ def allgather_rows(local, counts: List[int]):
     maxc = max(counts)
     if torch.is_tensor(local):
         t = local  # dtype preserved: fp32 from the GPU engine, fp64 CPU oracle
-        if dist.get_backend() == "nccl" and not t.is_cuda:
+        src_device = t.device
+        nccl = dist.get_backend() == "nccl"
+        if nccl and not t.is_cuda:
             t = t.cuda()
+        elif not nccl and t.is_cuda:
+            t = t.cpu()  # gloo transport (e.g. 2 ranks sharing one GPU)
         if t.shape[0] < maxc:  # pad ≤1 row (shard_bounds remainder)
             padrow = t.new_zeros((maxc - t.shape[0],) + tuple(t.shape[1:]))
             t = torch.cat([t, padrow], dim=0)
         t = t.contiguous()
         out = t.new_empty((world * maxc,) + tuple(t.shape[1:]))
         dist.all_gather_into_tensor(out, t)
-        if all(c == maxc for c in counts):
-            return out
-        blocks = [out[r * maxc : r * maxc + counts[r]] for r in range(world)]
-        return torch.cat(blocks, dim=0)
+        if not all(c == maxc for c in counts):
+            blocks = [out[r * maxc : r * maxc + counts[r]] for r in range(world)]
+            out = torch.cat(blocks, dim=0)
+        return out.to(src_device)
     device = "cuda" if dist.get_backend() == "nccl" else "cpu"
     tail = local.shape[1:]
     pad = np.zeros((maxc,) + tail, dtype=np.float64)
