@@ -164,7 +164,13 @@ def main() -> None:
 
     engine = KernelShapEngine(
         pred, background, groups=groups, link="logit", seed=0, device=device,
-        kernels=KernelConfig(predict_dtype=args.dtype),
+        kernels=KernelConfig(
+            predict_dtype=args.dtype,
+            # torch-module configs (mlp/resnet): bf16 dtype selects autocast
+            # around the module forward
+            module_autocast=("bf16" if args.dtype in ("bf16", "bf16x2")
+                             else "off"),
+        ),
     )
 
     if use_cuda:

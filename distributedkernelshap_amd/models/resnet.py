@@ -67,6 +67,12 @@ def resnet18(num_classes: int = 10, seed: int = 0):
             if x.dim() == 2:
                 hw = int((x.shape[1] // 3) ** 0.5)
                 x = x.view(-1, 3, hw, hw)
+            if self.stem[0].weight.is_contiguous(
+                memory_format=torch.channels_last
+            ):
+                # match the NHWC weight layout once at the input instead of
+                # per-conv internal transposes
+                x = x.contiguous(memory_format=torch.channels_last)
             x = self.stem(x)
             x = self.layers(x)
             x = self.pool(x).flatten(1)

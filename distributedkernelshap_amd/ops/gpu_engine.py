@@ -99,6 +99,15 @@ class GpuKernelShap:
             tm = getattr(engine.predictor, "torch_module", None)
             if tm is not None:
                 self.module = tm().to(self.device)
+                if engine.kernels.module_channels_last:
+                    # conv nets: NHWC weights select MIOpen's fast paths;
+                    # no-op for pure-linear modules
+                    try:
+                        self.module = self.module.to(
+                            memory_format=t.channels_last
+                        )
+                    except Exception:  # pragma: no cover
+                        pass
             else:
                 raise TypeError(
                     "GPU engine needs a predictor exposing linear_params() or "
@@ -173,7 +182,10 @@ class GpuKernelShap:
             return self.module(rows).double()
 
     def _predict_rows(self, rows):
-        """Run the predictor on a device tensor of rows -> (n, n_out) fp32."""
+        """Run the predictor on a device tensor of rows -> (n, n_out) fp32.
+        ``module_autocast='bf16'`` wraps the module forward in bf16 autocast
+        (MI355X bf16 matrix cores are 16x the f32 MFMA rate) — the lever for
+        the predict-bound mlp/resnet configs."""
         t = self.torch
         if self.linear is not None:
             z = rows @ self.linear["W"].T + self.linear["b"]
@@ -184,7 +196,11 @@ class GpuKernelShap:
                 return t.softmax(z, dim=-1)
             return z
         with t.no_grad():
-            out = self.module(rows)
+            if self.engine.kernels.module_autocast == "bf16":
+                with t.autocast(device_type="cuda", dtype=t.bfloat16):
+                    out = self.module(rows)
+            else:
+                out = self.module(rows)
         return out.float()
 
     def _x_part(self, X_dev):
