@@ -2,6 +2,11 @@
 test_engine_gpu_stress_paths): prints per-stage GPU-vs-fp64 error so a
 failure can be attributed to the tiled predict, the Gram build, or the
 solve conditioning."""
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
 import numpy as np
 import torch as t
 

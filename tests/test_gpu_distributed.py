@@ -98,20 +98,19 @@ def _sample_sharded_worker(rank, world, port, ret):
     )
     sv = explain_sample_sharded(eng, data.X)
     if rank == 0:
-        cpu = KernelShapEngine(
-            pred, data.background, groups=data.groups, link="logit", seed=0,
-            device="cpu",
-        )
-        ref = cpu.shap_values(data.X)
+        # single-GPU reference on the SAME device masks (counter RNG keyed by
+        # instance id) — a CPU oracle would draw different random coalitions
+        # and differ by sampling noise, not pipeline error
+        ref = eng.shap_values(data.X)
         ret.put(([s.copy() for s in sv], [s.copy() for s in ref]))
     dist.barrier()
     dist.destroy_process_group()
 
 
-def test_sample_sharded_gpu_matches_oracle():
+def test_sample_sharded_gpu_matches_single_gpu():
     """GPU sample-sharded mode (fused predict per nsamples slice + all-reduce
-    of the WLS normal equations) matches the CPU fp64 oracle to fp32
-    pipeline tolerance."""
+    of the WLS normal equations) matches the unsharded single-GPU result on
+    the same masks to fp32 pipeline tolerance."""
     sv, ref = _spawn2(_sample_sharded_worker, 29623)
     for o in range(2):
         err = np.abs(sv[o] - ref[o]).max()

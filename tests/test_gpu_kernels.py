@@ -88,6 +88,53 @@ def test_fill_random_masks_sharding_invariance(ext):
     assert torch.equal(full[2], single[0])
 
 
+def test_fill_random_masks_size_distribution(ext):
+    """The GPU sampler's subset-size draws follow the plan's residual
+    Shapley-kernel distribution (frequency test against the supplied CDF —
+    the CPU sampler has exact tests, this closes the loop for the device
+    Philox stream)."""
+    from distributedkernelshap_amd.core.sampler import plan_coalitions
+
+    m = 20
+    plan = plan_coalitions(m, 2 ** 14)
+    assert plan.n_random > 4000
+    b = 64
+    ne = plan.enum_masks.shape[0]
+    masks = torch.zeros(b, plan.nsamples, m, dtype=torch.uint8, device="cuda")
+    cdf = torch.tensor(
+        np.cumsum(plan.random_size_probs).astype(np.float32), device="cuda"
+    )
+    szs = torch.tensor(plan.random_sizes.astype(np.int32), device="cuda")
+    ids = torch.arange(b, dtype=torch.int32, device="cuda")
+    num_paired = int(np.floor((m - 1) / 2))
+    ext.fill_random_masks(
+        masks, ne, plan.n_random, cdf, szs, num_paired, 0, ids
+    )
+    sizes = masks[:, ne:].sum(dim=2).cpu().numpy().ravel()
+    n = sizes.size
+    # each draw of size s (prob p_s) emits one row of size s plus a
+    # complement row of size m-s when s <= num_paired
+    probs = {int(s): float(p) for s, p in
+             zip(plan.random_sizes, plan.random_size_probs)}
+    exp: dict = {}
+    for s, p in probs.items():
+        exp[s] = exp.get(s, 0.0) + p
+        if s <= num_paired:
+            exp[m - s] = exp.get(m - s, 0.0) + p
+    zsum = sum(exp.values())
+    for s in exp:
+        exp[s] /= zsum
+    # every emitted size must be an expected one
+    assert set(np.unique(sizes)).issubset(set(exp))
+    slack = 8 * b  # ≤1 truncated complement per fixed per-wave chunk
+    for s, p in exp.items():
+        cnt = int((sizes == s).sum())
+        sigma = (n * p * (1 - p)) ** 0.5
+        assert abs(cnt - n * p) < 6 * sigma + slack, (
+            s, cnt, n * p, sigma
+        )
+
+
 # ----------------------------------------------------------------------- #
 # K3-K6 fused predict
 
@@ -341,14 +388,18 @@ def test_engine_gpu_stress_paths():
     for o in range(2):
         total = sv[o].sum(axis=1) + eng.expected_value[o]
         assert np.abs(total - fx[:, o]).max() < 2e-3
-    # CPU fp64 oracle agreement (same masks via the counter RNG)
-    cpu = KernelShapEngine(
+    # fp64 device-mode agreement: SAME device masks (counter RNG), double
+    # arithmetic — isolates pipeline precision from sampling noise (a CPU
+    # oracle would draw different random coalitions)
+    from distributedkernelshap_amd.config import KernelConfig
+
+    eng64 = KernelShapEngine(
         pred, data.background, groups=data.groups, link="logit", seed=0,
-        device="cpu",
+        device="cuda", kernels=KernelConfig(predict_dtype="fp64"),
     )
-    sv_c = cpu.shap_values(data.X, nsamples=2048, l1_reg=False)
+    sv64 = eng64.shap_values(data.X, nsamples=2048, l1_reg=False)
     for o in range(2):
-        err = np.abs(sv[o] - sv_c[o]).max()
+        err = np.abs(sv[o] - sv64[o]).max()
         assert err < 2e-3, err
 
 
