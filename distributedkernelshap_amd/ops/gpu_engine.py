@@ -373,8 +373,7 @@ class GpuKernelShap:
         wbg = t.zeros(npad, device=self.device)
         wbg[: self.N] = self.bg_w
         n_ntiles = (npad + 127) // 128
-        nacc = 1 if act == 3 else self.n_out
-        partial = self._buf("ftpart", (b, n_ntiles, s, nacc))
+        partial = self._buf("ftpart", (b, n_ntiles, s, self.n_out))
         ey = self._buf("ey", (b, s, self.n_out))
         self.ext.fused_predict_tiled(masks, diff, base, wbg, partial, ey, act)
         return ey
@@ -811,7 +810,9 @@ class GpuKernelShap:
             if m <= 64:
                 # packed u64 masks feed the MFMA WLS Gram build
                 packed = self._buf("packed", (len(ids), plan.nsamples), t.int64)
+            pairwise = False
             if self.linear is not None:
+                act3 = self._act_oimg()[0] == 3
                 mpad = max(4, (m + 3) // 4 * 4)
                 npad = (self.N + 15) // 16 * 16
                 use_bf16 = (
@@ -826,22 +827,20 @@ class GpuKernelShap:
                         and self.n_out in (1, 2, 4)):
                     ey = self._ey_fused_linear(masks, sub_X, varying)
                 elif kc.fused_predict and self.n_out in (1, 2, 4):
-                    # stress shapes stay on the hand-written MFMA path
+                    # stress shapes stay on the hand-written MFMA path; the
+                    # tiled kernel dual-accumulates (p0, p1) so the pairwise
+                    # logit survives saturated probabilities
                     ey = self._ey_fused_tiled(masks, sub_X, varying)
+                    pairwise = act3
                 else:
                     ey = self._ey_linear_torch(masks, sub_X, varying)
+                    pairwise = act3  # torch softmax is relatively accurate
             else:
                 if packed is not None:
                     self.ext.pack_masks(masks, packed)
                 ey = self._ey_torch_module(masks, sub_X, varying)
             timer.mark("predict")
-            # in-place link transform: ey is a workspace, not needed afterwards
-            if self.link_name == "identity":
-                ey_adj = ey.sub_(lfnull[None, None, :])
-            else:
-                ey.clamp_(_EPS, 1.0 - _EPS)
-                ey.log_().sub_(t.log1p(-t.exp(ey)))  # log(p/(1-p)) in place
-                ey_adj = ey.sub_(lfnull[None, None, :])
+            ey_adj = self._link_ey(ey, lfnull, pairwise)
             total = total_all[ids_t].contiguous()
             if self._l1_active(plan, l1_reg):
                 if kc.l1_device:
@@ -911,6 +910,26 @@ class GpuKernelShap:
             return frac < 0.2
         return l1_reg not in (None, False, 0)
 
+    def _link_ey(self, ey, lfnull, pairwise):
+        """In-place link transform (ey is a workspace). ``pairwise``: ey
+        carries relatively-accurate (p0_sum, p1_sum) class sums, so the
+        logit is computed as log(p1) - log(p0) — the reference-faithful
+        fp64-numpy formula, with no clamp cliff at saturation."""
+        t = self.torch
+        if self.link_name == "identity":
+            return ey.sub_(lfnull[None, None, :])
+        if pairwise and self.n_out == 2:
+            tiny = 1e-300 if ey.dtype == t.float64 else 1e-38
+            lratio = (t.log(ey[..., 1].clamp_min(tiny))
+                      - t.log(ey[..., 0].clamp_min(tiny)))
+            ey[..., 0] = -lratio
+            ey[..., 1] = lratio
+            return ey.sub_(lfnull[None, None, :])
+        eps = 1e-15 if ey.dtype == t.float64 else _EPS
+        ey.clamp_(eps, 1.0 - eps)
+        ey.log_().sub_(t.log1p(-t.exp(ey)))  # log(p/(1-p)) in place
+        return ey.sub_(lfnull[None, None, :])
+
     def _bucket_fp64(self, plan, masks, sub_X, varying, ids_t, total_all,
                      lfnull64, l1_reg):
         """One bucket of the fp64 verification pipeline: fp64 ey on the same
@@ -933,8 +952,16 @@ class GpuKernelShap:
             ])
         ).to(self.device)
         kw = kw64[None, :].expand(b, plan.nsamples).contiguous()
+        act3 = self._act_oimg()[0] == 3
         if self.link_name == "identity":
             ey_adj = ey - lfnull64[None, None, :]
+        elif act3 and self.n_out == 2:
+            # pairwise logit on the fp64 softmax pair — same formula as the
+            # fp32 tiled path, so the self-check isolates arithmetic only
+            lratio = (t.log(ey[..., 1].clamp_min(1e-300))
+                      - t.log(ey[..., 0].clamp_min(1e-300)))
+            ey_adj = (t.stack([-lratio, lratio], dim=-1)
+                      - lfnull64[None, None, :])
         else:
             p = ey.clamp(1e-15, 1.0 - 1e-15)
             ey_adj = t.log(p / (1.0 - p)) - lfnull64[None, None, :]

@@ -242,7 +242,7 @@ void fused_predict_tiled(
     int Mpad = diff.size(2), Npad = diff.size(3);
     int n_out = ey.size(2);
     int oimg = (act == 3) ? 1 : n_out;
-    int nacc = (act == 3) ? 1 : n_out;
+    int nacc = n_out;   // act 3 dual-accumulates p0 and p1
     int n_ntiles = (Npad + 127) / 128;
     TORCH_CHECK(diff.size(1) == oimg, "diff image count vs act");
     TORCH_CHECK(M <= Mpad, "M vs Mpad");

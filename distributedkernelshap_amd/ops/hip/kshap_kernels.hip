@@ -1364,8 +1364,14 @@ void fused_predict_tiled_kernel(
     float* __restrict__ partial_out,    // (B, n_ntiles, S, NACC)
     int B, int S, int M, int Mpad, int Npad)
 {
+    // ACT 3 (binary softmax from the logit difference) accumulates BOTH
+    // class sums here — p0 = p1*exp(-z) keeps RELATIVE accuracy for
+    // near-saturated probabilities, so the engine's pairwise
+    // log(p1)-log(p0) link stays accurate where a (1 - p1) complement
+    // would lose everything to fp32 cancellation (stress configs saturate
+    // their logits; the reference's fp64 numpy has no such cliff)
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
-    constexpr int NACC = (ACT == 3) ? 1 : NOUT;
+    constexpr int NACC = NOUT;
     constexpr int KC = (OIMG >= 4) ? 16 : 32;   // LDS diff chunk k-depth
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
@@ -1460,14 +1466,18 @@ void fused_predict_tiled_kernel(
                     z[o] = acc[ct][o][r] + base_lds[o * 128 + n];
                 float zz[NACC];
                 if (ACT == 3) {
-                    zz[0] = fast_rcp(1.0f + __expf(-z[0]));
+                    const float e = __expf(-z[0]);
+                    const float p1 = fast_rcp(1.0f + e);
+                    zz[1] = p1;
+                    zz[0] = p1 * e;        // sigma(-z): exact-complement sum
                 } else if (ACT == 1) {
 #pragma unroll
                     for (int o = 0; o < NOUT; ++o)
                         zz[o] = fast_rcp(1.0f + __expf(-z[o]));
                 } else if (ACT == 2 && NOUT == 2) {
-                    const float p1 = fast_rcp(1.0f + __expf(z[0] - z[1]));
-                    zz[0] = 1.0f - p1;
+                    const float e = __expf(z[0] - z[1]);
+                    const float p1 = fast_rcp(1.0f + e);
+                    zz[0] = p1 * e;
                     zz[1] = p1;
                 } else if (ACT == 2) {
                     float mx = z[0];
@@ -1521,7 +1531,7 @@ void fused_predict_tiled_kernel(
 __global__ void reduce_partials_kernel(
     const float* __restrict__ partial,  // (B, n_ntiles, S, nacc)
     float* __restrict__ ey,             // (B, S, n_out)
-    size_t total_rows, int S, int n_ntiles, int nacc, int n_out, int act)
+    size_t total_rows, int S, int n_ntiles, int nacc, int n_out)
 {
     const size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (idx >= total_rows) return;       // total_rows = B * S
@@ -1531,12 +1541,7 @@ __global__ void reduce_partials_kernel(
         float v = 0.0f;
         for (int nt = 0; nt < n_ntiles; ++nt)
             v += partial[(((size_t)b * n_ntiles + nt) * S + s) * nacc + o];
-        if (act == 3) {
-            ey[((size_t)b * S + s) * n_out + 0] = 1.0f - v;
-            ey[((size_t)b * S + s) * n_out + n_out - 1] = v;
-        } else {
-            ey[((size_t)b * S + s) * n_out + o] = v;
-        }
+        ey[((size_t)b * S + s) * n_out + o] = v;
     }
 }
 
@@ -1547,7 +1552,7 @@ static void launch_ft(
     int Mpad, int Npad, hipStream_t stream)
 {
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
-    constexpr int NACC = (ACT == 3) ? 1 : NOUT;
+    constexpr int NACC = NOUT;
     constexpr int KC = (OIMG >= 4) ? 16 : 32;
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
@@ -1558,7 +1563,7 @@ static void launch_ft(
     size_t rows = (size_t)B * S;
     reduce_partials_kernel<<<dim3((unsigned)((rows + 255) / 256)), dim3(256),
                              0, stream>>>(
-        partial, ey, rows, S, n_ntiles, NACC, NOUT, ACT);
+        partial, ey, rows, S, n_ntiles, NACC, NOUT);
 }
 
 template <int NOUT>
