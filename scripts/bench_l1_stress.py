@@ -39,3 +39,45 @@ def timed(label, **kw):
 t_off = timed("l1_reg=False  ", l1_reg=False)
 t_auto = timed("l1_reg='auto' ", l1_reg="auto")
 print(f"auto/off ratio: {t_auto/t_off:.2f}x")
+
+# where should the LARS path iteration live? time the select alone on GPU
+# tensors vs CPU copies of the same normal equations
+import torch
+
+from distributedkernelshap_amd.core.lars import batched_lars_select
+
+gpu = eng._gpu
+X_dev = torch.tensor(data.X, dtype=torch.float32, device="cuda")
+varying = np.arange(256)
+plan = eng._plan(256, 2 ** 14)
+masks, kw = gpu._device_masks(plan, np.arange(64))
+ey = gpu._ey_fused_tiled(masks, X_dev, varying)
+lfnull = gpu._link(gpu.fnull.double()).float()
+ey_adj = gpu._link_ey(ey, lfnull, True)
+for where in ("cuda", "cpu"):
+    sup = None
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    support = gpu._l1_select_batched(masks, kw, ey_adj, "auto")
+    torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    print(f"_l1_select_batched (path on {where}): {(t1-t0)*1e3:.0f} ms")
+    if where == "cuda":
+        # monkeypatch: run the path on CPU copies next iteration
+        orig = batched_lars_select
+
+        def cpu_select(G, c, yty, **kwargs):
+            zb = kwargs.pop("zbar", None)
+            yb = kwargs.pop("ybar", None)
+            out = orig(
+                G.cpu(), c.cpu(), yty.cpu(),
+                zbar=None if zb is None else zb.cpu(),
+                ybar=None if yb is None else yb.cpu(), **kwargs,
+            )
+            return out.to(G.device)
+
+        import distributedkernelshap_amd.ops.gpu_engine as ge
+
+        sys.modules["distributedkernelshap_amd.core.lars"].batched_lars_select = cpu_select
+        # gpu_engine imports it inside the method, so the module attr is used
+
