@@ -112,13 +112,16 @@ def batched_lars_select(
 
     for _step in range(max_steps):
         live = ~frozen
-        if not bool(live.any()):
-            break
-
         dropping = drop_pending & live
         appending = live & ~dropping
+        # ONE host sync per step: all control flags in a single transfer
+        flags = t.stack([
+            live.any(), dropping.any(), appending.any()
+        ]).cpu()
+        if not bool(flags[0]):
+            break
 
-        if bool(dropping.any()):
+        if bool(flags[1]):
             # remove slot p: permute it to the last active slot, then reduce
             # K by the reverse Schur complement  K11 - k1 k1^T / k22
             di = t.nonzero(dropping).squeeze(1)
@@ -157,14 +160,12 @@ def batched_lars_select(
         frozen = frozen | stop_now
         appending = appending & ~stop_now
         live = ~frozen
-        if not bool(live.any()):
-            break
 
-        # all slot-space work is confined to the leading kcap slots (k grows
-        # in lockstep across instances, so this caps the K-update cost at
-        # O(b * k^2) per step instead of O(b * m^2))
-        kcap = min(int(k.max().item()) + 1, m)
-        if bool(appending.any()):
+        # all slot-space work is confined to the leading kcap slots. k grows
+        # by at most 1 per step, so min(step+1, m) bounds every live count
+        # WITHOUT a device sync, and the padded slots are exact zeros.
+        kcap = min(_step + 1, m)
+        if bool(flags[2]):
             app = appending
             gj = t.gather(
                 G, 2, jstar.view(b, 1, 1).expand(b, m, 1)

@@ -844,8 +844,12 @@ class GpuKernelShap:
             total = total_all[ids_t].contiguous()
             if self._l1_active(plan, l1_reg):
                 if kc.l1_device:
-                    support = self._l1_select_batched(masks, kw, ey_adj, l1_reg)
-                    phi = self._solve_selected(masks, kw, ey_adj, total, support)
+                    support, g64, r64 = self._l1_select_batched(
+                        masks, kw, ey_adj, l1_reg
+                    )
+                    phi = self._solve_selected(
+                        masks, kw, ey_adj, total, support, g64, r64
+                    )
                 else:
                     phi = self._solve_host_l1(masks, kw, ey_adj, total, l1_reg)
             else:
@@ -1072,15 +1076,23 @@ class GpuKernelShap:
                 ).squeeze(1)
             zbar = zsum / s
             ybar = (sqw.double() * y.double()).sum(1) / s
-        return batched_lars_select(
+        support = batched_lars_select(
             g64, c64, yty, n_samples=s, mode=mode, num_features=nf,
             alpha=alpha, zbar=zbar, ybar=ybar,
         )
+        return support, g64, r64
 
-    def _solve_selected(self, masks, kw, ey_adj, total, support):
-        """Constrained WLS over per-instance selected supports: instances are
-        grouped by identical support (the common case) and each group solved
-        with the normal kernel dispatch on the column-sliced masks."""
+    def _solve_selected(self, masks, kw, ey_adj, total, support, g64, r64):
+        """Constrained WLS over per-instance selected supports, derived
+        ALGEBRAICALLY from the full raw normal equations the selection
+        already built (no second pass over the S samples): with s = selected
+        features, l = last of them,
+
+          A_el[i,j] = G[si,sj] - G[si,l] - G[l,sj] + G[l,l]
+          r_el[i,o] = (c_o[si] - c_o[l]) - (G[si,l] - G[l,l]) * total_o
+
+        Instances are grouped by identical support and each group solved as
+        one batched fp64 system."""
         t = self.torch
         b, s, m = masks.shape
         phi = t.zeros(b, m, self.n_out, device=self.device)
@@ -1090,43 +1102,31 @@ class GpuKernelShap:
             groups.setdefault(sup_np[i].tobytes(), []).append(i)
         for idxs in groups.values():
             sel = np.nonzero(sup_np[idxs[0]])[0]
-            ids_t = t.tensor(idxs, dtype=t.int64, device=self.device)
-            tot_g = total[ids_t].contiguous()
             if len(sel) == 0:
-                # empty selection: shap solves over all features
-                phi[ids_t] = self._solve_for(
-                    masks[ids_t].contiguous(), kw[ids_t].contiguous(),
-                    ey_adj[ids_t].contiguous(), tot_g,
-                )
-                continue
+                sel = np.arange(m)       # empty selection: solve over all
+            ids_t = t.tensor(idxs, dtype=t.int64, device=self.device)
+            tot_g = total[ids_t].double()                  # (bg, n_out)
             if len(sel) == 1:
-                phi[ids_t, int(sel[0])] = tot_g
+                phi[ids_t, int(sel[0])] = tot_g.float()
                 continue
             sel_t = t.tensor(sel, dtype=t.int64, device=self.device)
-            sub = masks[ids_t][:, :, sel_t].contiguous()
-            phi_sub = self._solve_for(
-                sub, kw[ids_t].contiguous(), ey_adj[ids_t].contiguous(), tot_g
-            )
+            st, l = sel_t[:-1], int(sel_t[-1])
+            gg = g64[ids_t]                                # (bg, m, m)
+            rr = r64[ids_t]                                # (bg, m, n_out)
+            gss = gg[:, st][:, :, st]
+            gcol = gg[:, st, l]                            # (bg, k-1)
+            gll = gg[:, l, l]
+            a_el = gss - gcol[:, :, None] - gcol[:, None, :] + gll[:, None, None]
+            r_el = ((rr[:, st] - rr[:, l][:, None, :])
+                    - (gcol - gll[:, None])[:, :, None] * tot_g[:, None, :])
+            try:
+                w = t.linalg.solve(a_el, r_el)
+            except Exception:
+                w = t.linalg.lstsq(a_el, r_el).solution
+            phi_last = tot_g[:, None, :] - w.sum(dim=1, keepdim=True)
+            phi_sub = t.cat([w, phi_last], dim=1).float()
             phi[ids_t[:, None], sel_t[None, :]] = phi_sub
         return phi
-
-    def _solve_for(self, masks, kw, ey_adj, total):
-        """WLS solve dispatch for arbitrary (sub-)shapes."""
-        t = self.torch
-        b, s, m = masks.shape
-        kc = self.engine.kernels
-        if (kc.wls_mode != "torch" and 2 <= m <= 64 and self.n_out <= 8
-                and (m - 1 + self.n_out <= 16 or m <= 24)):
-            phi = t.empty(b, m, self.n_out, device=self.device)
-            packed = None
-            if m - 1 + self.n_out <= 16 and kc.wls_mode != "generic":
-                packed = t.empty(b, s, dtype=t.int64, device=self.device)
-                self.ext.pack_masks(masks, packed)
-            self.ext.wls_solve(masks, kw, ey_adj, total, phi, packed)
-            return phi
-        if kc.wls_mode in ("auto", "mfma") and m <= 513 and self.n_out <= 8:
-            return self._solve_gram(masks, kw, ey_adj, total)
-        return self._solve_torch(masks, kw, ey_adj, total)
 
     def _solve_host_l1(self, masks, kw, ey_adj, total, l1_reg):
         """Cold path: l1 feature selection + solve on host, per instance."""
