@@ -604,8 +604,17 @@ class GpuKernelShap:
         from ..core.sampler import default_nsamples
 
         s = nsamples or default_nsamples(max(2, self.n_groups))
-        # masks u8 + diff images + packed u64 + kwb f32 + ey f32 per instance
-        per_inst = s * (self.n_groups + 8 + 4 + 4 * self.n_out) + (1 << 14)
+        g = self.n_groups
+        # masks u8 + packed u64 + kwb f32 + ey f32 (+ column-tile partials
+        # on the tiled stress path) per instance
+        npad = (self.N + 15) // 16 * 16
+        n_ntiles = (npad + 127) // 128
+        per_inst = s * (g + 8 + 4 + 4 * self.n_out) + (1 << 14)
+        if g > 64 or npad > 128:
+            per_inst += s * 4 * self.n_out * n_ntiles        # ftpart
+            mpad = max(4, (g + 3) // 4 * 4)
+            per_inst += 4 * mpad * npad * max(1, self.n_out) # diff image
+            per_inst += 8 * g * (g + self.n_out) * 2         # Gram + LU work
         return max(1, self._CHUNK_BYTES // per_inst)
 
     def shap_values(
