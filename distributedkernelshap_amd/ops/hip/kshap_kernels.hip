@@ -162,12 +162,35 @@ __global__ void fill_random_masks_kernel(
                 ++guard;
             }
             uint8_t* row = mrow_base + (size_t)(ne + written + rows_before) * M;
-            for (int g = 0; g < M; ++g)
-                row[g] = (uint8_t)((bits[g >> 6] >> (g & 63)) & 1ull);
+            const bool wide = (M & 3) == 0;   // rows 4-aligned: dword stores
+            if (wide) {
+                for (int g = 0; g < M; g += 4) {
+                    const uint32_t nib =
+                        (uint32_t)((bits[g >> 6] >> (g & 63)) & 0xFull);
+                    // bit i of the nibble -> byte i (mask bytes are 0/1)
+                    *(uint32_t*)(row + g) =
+                        (nib & 1u) | ((nib & 2u) << 7) | ((nib & 4u) << 14)
+                        | ((nib & 8u) << 21);
+                }
+            } else {
+                for (int g = 0; g < M; ++g)
+                    row[g] = (uint8_t)((bits[g >> 6] >> (g & 63)) & 1ull);
+            }
             if (paired && rows_before + 1 < remaining) {
                 uint8_t* crow = row + M;
-                for (int g = 0; g < M; ++g)
-                    crow[g] = (uint8_t)(1u - ((bits[g >> 6] >> (g & 63)) & 1ull));
+                if (wide) {
+                    for (int g = 0; g < M; g += 4) {
+                        const uint32_t nib = (uint32_t)(
+                            (~bits[g >> 6] >> (g & 63)) & 0xFull);
+                        *(uint32_t*)(crow + g) =
+                            (nib & 1u) | ((nib & 2u) << 7) | ((nib & 4u) << 14)
+                            | ((nib & 8u) << 21);
+                    }
+                } else {
+                    for (int g = 0; g < M; ++g)
+                        crow[g] = (uint8_t)(
+                            1u - ((bits[g >> 6] >> (g & 63)) & 1ull));
+                }
             }
         }
         int consumed = total_rows < remaining ? total_rows : remaining;
@@ -1135,34 +1158,39 @@ extern "C" int launch_wls_solve(
 __global__ void pack_masks_words_kernel(
     const uint8_t* __restrict__ masks,  // (B, S, M)
     uint64_t* __restrict__ packed,      // (B, S, W)
-    size_t n_rows, int M, int W)
+    size_t n_items, int M, int W)
 {
-    const size_t row = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (row >= n_rows) return;
+    // one thread per OUTPUT WORD (row, w): each thread reads its 64-byte
+    // source span — exactly one cache line when M%64==0 — and consecutive
+    // threads (w fastest) touch consecutive lines, so both the mask reads
+    // and the packed writes coalesce (the thread-per-row variant issued
+    // 256-byte-strided dword loads: 16x line amplification at M=256)
+    const size_t item = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (item >= n_items) return;
+    const size_t row = item / W;
+    const int w = (int)(item % W);
     const uint8_t* src = masks + row * M;
-    for (int w = 0; w < W; ++w) {
-        const int k0 = w * 64;
-        const int k1 = min(M, k0 + 64);
-        uint64_t bits = 0ull;
-        int k = k0;
-        if ((((size_t)(src + k0)) & 3) == 0) {
-            for (; k + 4 <= k1; k += 4) {
-                uint32_t word = *(const uint32_t*)(src + k);
+    const int k0 = w * 64;
+    const int k1 = min(M, k0 + 64);
+    uint64_t bits = 0ull;
+    int k = k0;
+    if ((((size_t)(src + k0)) & 3) == 0) {
+        for (; k + 4 <= k1; k += 4) {
+            uint32_t word = *(const uint32_t*)(src + k);
 #pragma unroll
-                for (int j = 0; j < 4; ++j)
-                    bits |= ((uint64_t)((word >> (8 * j)) & 1u)) << (k - k0 + j);
-            }
+            for (int j = 0; j < 4; ++j)
+                bits |= ((uint64_t)((word >> (8 * j)) & 1u)) << (k - k0 + j);
         }
-        for (; k < k1; ++k) bits |= ((uint64_t)(src[k] & 1)) << (k - k0);
-        packed[row * W + w] = bits;
     }
+    for (; k < k1; ++k) bits |= ((uint64_t)(src[k] & 1)) << (k - k0);
+    packed[row * W + w] = bits;
 }
 
 extern "C" void launch_pack_masks_words(
     const uint8_t* masks, uint64_t* packed, int B, int S, int M, int W,
     hipStream_t stream)
 {
-    size_t n = (size_t)B * S;
+    size_t n = (size_t)B * S * W;
     pack_masks_words_kernel<<<dim3((unsigned)((n + 255) / 256)), dim3(256), 0,
                               stream>>>(masks, packed, n, M, W);
 }
@@ -1198,13 +1226,24 @@ void wls_gram_kernel(
     const float* __restrict__ total,      // (B, n_out)
     double* __restrict__ A64,             // (B, mm, mm)
     double* __restrict__ rhs64,           // (B, mm, n_out)
-    int B, int S, int M, int W, int n_out, int TI)
+    int B, int S, int M, int W, int n_out, int NU)
 {
-    const int b = blockIdx.x / TI;
-    const int ti = blockIdx.x % TI;
+    // one workgroup per (instance, tj-tile-block) work UNIT: a strip-per-ti
+    // grid left the chip load-imbalanced (ti=0 owns 4x the tiles of the
+    // last strip) and under-filled at 1024 workgroups
+    const int b = blockIdx.x / NU;
+    int u = blockIdx.x % NU;
     const int mm = M - 1;
     const int cols = mm + n_out;
     const int TJ = (cols + 15) / 16;
+    const int TI = (mm + 15) / 16;
+    int ti = 0;
+    for (; ti < TI; ++ti) {
+        const int blocks = (TJ - ti + GRAM_TJB - 1) / GRAM_TJB;
+        if (u < blocks) break;
+        u -= blocks;
+    }
+    const int tj0 = ti + u * GRAM_TJB;
     const int tid = threadIdx.x;
     const int lane = tid & (WAVE - 1);
     const int wv = tid >> 6;
@@ -1229,7 +1268,7 @@ void wls_gram_kernel(
     const int wi = i >> 6, ibit = i & 63;
     const int lw = (M - 1) >> 6, lb = (M - 1) & 63;  // last-mask bit coords
 
-    for (int tj0 = ti; tj0 < TJ; tj0 += GRAM_TJB) {
+    {
         const int ntj = min(GRAM_TJB, TJ - tj0);
         double dacc[GRAM_TJB][4];
         f32x4 acc[GRAM_TJB];
@@ -1333,10 +1372,14 @@ extern "C" int launch_wls_gram(
         return -1;
     const int mm = M - 1;
     const int TI = (mm + 15) / 16;
+    const int TJ = (mm + n_out + 15) / 16;
+    int NU = 0;   // upper-triangle tile blocks (rhs cols ride the last ones)
+    for (int ti = 0; ti < TI; ++ti)
+        NU += (TJ - ti + GRAM_TJB - 1) / GRAM_TJB;
     size_t lds = (size_t)GRAM_CHUNK * W * 8 + GRAM_CHUNK * 4
                + (size_t)GRAM_CHUNK * n_out * 4 + (size_t)4 * 16 * 17 * 8;
-    wls_gram_kernel<<<dim3(B * TI), dim3(256), lds, stream>>>(
-        packed, kw, ey_adj, total, A64, rhs64, B, S, M, W, n_out, TI);
+    wls_gram_kernel<<<dim3(B * NU), dim3(256), lds, stream>>>(
+        packed, kw, ey_adj, total, A64, rhs64, B, S, M, W, n_out, NU);
     return 0;
 }
 
