@@ -1447,27 +1447,18 @@ void fused_predict_tiled_kernel(
     const uint8_t* mlane = masksU + ((size_t)b * S + swave + arow) * M;
     const float* dlane = diff_lds + akcol * FT_NSTRIDE + arow;
 
-    // TWO s-sub-tiles (2 x 64 rows) per pass: every staged LDS value feeds
-    // two MFMAs (register blocking on the A side) and the per-chunk
-    // staging/barriers amortise over 128 rows instead of 64
-    for (int sub = 0; sub < S_TILE / S_SUB; sub += 2) {
+    for (int sub = 0; sub < S_TILE / S_SUB; ++sub) {
         const int ssub0 = s0 + sub * S_SUB;
         if (ssub0 >= S) break;
-        const int srow0 = ssub0 + swave + arow;
-        const int srow1 = srow0 + S_SUB;
-        const bool sv0 = srow0 < S;
-        const bool sv1 = srow1 < S;
-        const uint8_t* mrow0 = mlane + (size_t)ssub0 * M;
-        const uint8_t* mrow1 = mrow0 + (size_t)S_SUB * M;
+        const int srow = ssub0 + swave + arow;
+        const bool svalid = srow < S;
+        const uint8_t* mrow = mlane + (size_t)ssub0 * M;
 
-        f32x4 acc[2][FT_NTILE][OIMG];
+        f32x4 acc[FT_NTILE][OIMG];
 #pragma unroll
-        for (int h = 0; h < 2; ++h)
+        for (int ct = 0; ct < FT_NTILE; ++ct)
 #pragma unroll
-            for (int ct = 0; ct < FT_NTILE; ++ct)
-#pragma unroll
-                for (int o = 0; o < OIMG; ++o)
-                    acc[h][ct][o] = (f32x4){0, 0, 0, 0};
+            for (int o = 0; o < OIMG; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
 
         for (int kc0 = 0; kc0 < Mpad; kc0 += KC) {
             __syncthreads();
@@ -1485,101 +1476,93 @@ void fused_predict_tiled_kernel(
             const int kend = min(KC, Mpad - kc0);
             for (int ks = 0; ks < kend; ks += 4) {
                 const int k = kc0 + ks + akcol;
-                const bool kv = k < M;
-                const float a0 = (sv0 && kv) ? (float)(mrow0[k] & 1) : 0.0f;
-                const float a1 = (sv1 && kv) ? (float)(mrow1[k] & 1) : 0.0f;
+                const float a =
+                    (svalid && k < M) ? (float)(mrow[k] & 1) : 0.0f;
 #pragma unroll
                 for (int ct = 0; ct < FT_NTILE; ++ct)
 #pragma unroll
                     for (int o = 0; o < OIMG; ++o) {
                         const float bv =
                             dlane[(o * KC + ks) * FT_NSTRIDE + ct * 16];
-                        acc[0][ct][o] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                            a0, bv, acc[0][ct][o], 0, 0, 0);
-                        acc[1][ct][o] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                            a1, bv, acc[1][ct][o], 0, 0, 0);
+                        acc[ct][o] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            a, bv, acc[ct][o], 0, 0, 0);
                     }
             }
         }
 
         // epilogue: activation + weighted partial reduction over THIS
         // column tile (cols beyond ncols carry wbg 0 and contribute nothing)
+        float partialv[NACC][4];
 #pragma unroll
-        for (int h = 0; h < 2; ++h) {
-            if (h == 1 && ssub0 + S_SUB >= S) break;
-            float partialv[NACC][4];
+        for (int o = 0; o < NACC; ++o)
 #pragma unroll
-            for (int o = 0; o < NACC; ++o)
+            for (int r = 0; r < 4; ++r) partialv[o][r] = 0.0f;
 #pragma unroll
-                for (int r = 0; r < 4; ++r) partialv[o][r] = 0.0f;
+        for (int ct = 0; ct < FT_NTILE; ++ct) {
+            const int n = ct * 16 + arow;
+            const float wn = wbg_lds[n];
 #pragma unroll
-            for (int ct = 0; ct < FT_NTILE; ++ct) {
-                const int n = ct * 16 + arow;
-                const float wn = wbg_lds[n];
+            for (int r = 0; r < 4; ++r) {
+                float z[OIMG];
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    float z[OIMG];
+                for (int o = 0; o < OIMG; ++o)
+                    z[o] = acc[ct][o][r] + base_lds[o * 128 + n];
+                float zz[NACC];
+                if (ACT == 3) {
+                    const float e = __expf(-z[0]);
+                    const float p1 = fast_rcp(1.0f + e);
+                    zz[1] = p1;
+                    zz[0] = p1 * e;        // sigma(-z): exact-complement sum
+                } else if (ACT == 1) {
 #pragma unroll
-                    for (int o = 0; o < OIMG; ++o)
-                        z[o] = acc[h][ct][o][r] + base_lds[o * 128 + n];
-                    float zz[NACC];
-                    if (ACT == 3) {
-                        const float e = __expf(-z[0]);
-                        const float p1 = fast_rcp(1.0f + e);
-                        zz[1] = p1;
-                        zz[0] = p1 * e;    // sigma(-z): exact-complement sum
-                    } else if (ACT == 1) {
+                    for (int o = 0; o < NOUT; ++o)
+                        zz[o] = fast_rcp(1.0f + __expf(-z[o]));
+                } else if (ACT == 2 && NOUT == 2) {
+                    const float e = __expf(z[0] - z[1]);
+                    const float p1 = fast_rcp(1.0f + e);
+                    zz[0] = p1 * e;
+                    zz[1] = p1;
+                } else if (ACT == 2) {
+                    float mx = z[0];
 #pragma unroll
-                        for (int o = 0; o < NOUT; ++o)
-                            zz[o] = fast_rcp(1.0f + __expf(-z[o]));
-                    } else if (ACT == 2 && NOUT == 2) {
-                        const float e = __expf(z[0] - z[1]);
-                        const float p1 = fast_rcp(1.0f + e);
-                        zz[0] = p1 * e;
-                        zz[1] = p1;
-                    } else if (ACT == 2) {
-                        float mx = z[0];
+                    for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
+                    float sum = 0.0f;
 #pragma unroll
-                        for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
-                        float sum = 0.0f;
-#pragma unroll
-                        for (int o = 0; o < NOUT; ++o) {
-                            zz[o] = __expf(z[o] - mx);
-                            sum += zz[o];
-                        }
-                        const float inv = fast_rcp(sum);
-#pragma unroll
-                        for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
-                    } else {
-#pragma unroll
-                        for (int o = 0; o < NACC; ++o) zz[o] = z[o];
+                    for (int o = 0; o < NOUT; ++o) {
+                        zz[o] = __expf(z[o] - mx);
+                        sum += zz[o];
                     }
+                    const float inv = fast_rcp(sum);
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
+                } else {
+#pragma unroll
+                    for (int o = 0; o < NACC; ++o) zz[o] = z[o];
+                }
+#pragma unroll
+                for (int o = 0; o < NACC; ++o) partialv[o][r] += wn * zz[o];
+            }
+        }
+#pragma unroll
+        for (int o = 0; o < NACC; ++o)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float v = partialv[o][r];
+                v += __shfl_xor(v, 1);
+                v += __shfl_xor(v, 2);
+                v += __shfl_xor(v, 4);
+                v += __shfl_xor(v, 8);
+                partialv[o][r] = v;
+            }
+        if (arow == 0) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int ss = ssub0 + swave + akcol * 4 + r;
+                if (ss < S)
 #pragma unroll
                     for (int o = 0; o < NACC; ++o)
-                        partialv[o][r] += wn * zz[o];
-                }
-            }
-#pragma unroll
-            for (int o = 0; o < NACC; ++o)
-#pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    float v = partialv[o][r];
-                    v += __shfl_xor(v, 1);
-                    v += __shfl_xor(v, 2);
-                    v += __shfl_xor(v, 4);
-                    v += __shfl_xor(v, 8);
-                    partialv[o][r] = v;
-                }
-            if (arow == 0) {
-#pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    const int ss = ssub0 + h * S_SUB + swave + akcol * 4 + r;
-                    if (ss < S)
-#pragma unroll
-                        for (int o = 0; o < NACC; ++o)
-                            partial_out[(((size_t)b * n_ntiles + nt) * S + ss)
-                                            * NACC + o] = partialv[o][r];
-                }
+                        partial_out[(((size_t)b * n_ntiles + nt) * S + ss)
+                                        * NACC + o] = partialv[o][r];
             }
         }
     }
