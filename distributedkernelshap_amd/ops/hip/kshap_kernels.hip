@@ -1415,7 +1415,9 @@ void fused_predict_tiled_kernel(
     // their logits; the reference's fp64 numpy has no such cliff)
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
     constexpr int NACC = NOUT;
-    constexpr int KC = (OIMG >= 4) ? 16 : 32;   // LDS diff chunk k-depth
+    // 16-deep k chunks keep the LDS footprint ~10 KB -> 6 workgroups/CU
+    // (32-deep measured slower: 19 KB capped occupancy at 3/CU)
+    constexpr int KC = 16;                      // LDS diff chunk k-depth
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     const int stile = blockIdx.x % n_stiles;
@@ -1596,7 +1598,7 @@ static void launch_ft(
 {
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
     constexpr int NACC = NOUT;
-    constexpr int KC = (OIMG >= 4) ? 16 : 32;
+    constexpr int KC = 16;
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     dim3 grid(B * n_ntiles * n_stiles), block(256);
