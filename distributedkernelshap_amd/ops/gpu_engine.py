@@ -359,7 +359,9 @@ class GpuKernelShap:
         by a deterministic second kernel (no float atomics)."""
         t = self.torch
         b, s, m = masks.shape
-        mpad = max(4, (m + 3) // 4 * 4)
+        # Mpad rounds to the kernel's 16-deep LDS k-chunks so the inner MFMA
+        # loop is fully unrolled (pad k-slices are zeros in the diff image)
+        mpad = max(16, (m + 15) // 16 * 16)
         npad = (self.N + 15) // 16 * 16
         if vidx_t is None:
             vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)

@@ -1253,7 +1253,9 @@ void wls_gram_kernel(
     extern __shared__ char smem[];
     uint64_t* pk = (uint64_t*)smem;                       // GRAM_CHUNK * W
     float* wch = (float*)(pk + GRAM_CHUNK * W);           // GRAM_CHUNK
-    float* eych = wch + GRAM_CHUNK;                       // GRAM_CHUNK * n_out
+    float* mlch = wch + GRAM_CHUNK;                       // GRAM_CHUNK
+    float* wmlch = mlch + GRAM_CHUNK;                     // GRAM_CHUNK
+    float* eych = wmlch + GRAM_CHUNK;                     // GRAM_CHUNK * n_out
     double* red = (double*)(eych + GRAM_CHUNK * n_out);   // 4 * 16 * 17
     __shared__ float tot_s[WLS_MAX_NOUT];
 
@@ -1267,6 +1269,19 @@ void wls_gram_kernel(
     const int i = ti * 16 + arow;          // this lane's Gram row (A operand)
     const int wi = i >> 6, ibit = i & 63;
     const int lw = (M - 1) >> 6, lb = (M - 1) & 63;  // last-mask bit coords
+    // per-lane B-column descriptors — loop-invariant across the whole S
+    // scan (the in-loop recompute cost 36 VALU per MFMA, PMC-measured)
+    int jw[GRAM_TJB], jb[GRAM_TJB], jo[GRAM_TJB];
+    bool jgram[GRAM_TJB], jrhs[GRAM_TJB];
+#pragma unroll
+    for (int tt = 0; tt < GRAM_TJB; ++tt) {
+        const int j = (tj0 + tt) * 16 + arow;
+        jgram[tt] = j < mm;
+        jrhs[tt] = (j >= mm) && (j < cols);
+        jw[tt] = (j < mm) ? (j >> 6) : 0;
+        jb[tt] = j & 63;
+        jo[tt] = j - mm;
+    }
 
     {
         const int ntj = min(GRAM_TJB, TJ - tj0);
@@ -1285,12 +1300,17 @@ void wls_gram_kernel(
             for (int idx = tid; idx < clen * W; idx += 256)
                 pk[idx] = pbase[(size_t)(c0 + idx / W) * W + (idx % W)];
             for (int idx = tid; idx < clen; idx += 256) {
-                wch[idx] = kwb[c0 + idx];
+                const float w = kwb[c0 + idx];
                 const uint64_t lastw = pbase[(size_t)(c0 + idx) * W + lw];
                 const float ml = (float)((lastw >> lb) & 1ull);
+                wch[idx] = w;
+                mlch[idx] = ml;
+                wmlch[idx] = w * ml;
+                // rhs columns staged pre-weighted: bv is then a pure load
                 for (int o = 0; o < n_out; ++o)
                     eych[idx * n_out + o] =
-                        eyb[(size_t)(c0 + idx) * n_out + o] - ml * tot_s[o];
+                        w * (eyb[(size_t)(c0 + idx) * n_out + o]
+                             - ml * tot_s[o]);
             }
             __syncthreads();
             // wave wv covers samples [wv*64, wv*64+64) of the chunk; the
@@ -1300,23 +1320,25 @@ void wls_gram_kernel(
             for (int t4 = wlo; t4 < whi; t4 += 4) {
                 const int t = t4 + akk;
                 const bool tv = t < whi;
-                float a = 0.0f, ml = 0.0f, w = 0.0f;
+                float a = 0.0f, w = 0.0f, wml = 0.0f;
                 const uint64_t* bits = pk + (size_t)t * W;
                 if (tv) {
-                    ml = (float)((bits[lw] >> lb) & 1ull);
                     w = wch[t];
-                    if (i < mm) a = (float)((bits[wi] >> ibit) & 1ull) - ml;
+                    wml = wmlch[t];
+                    if (i < mm)
+                        a = (((bits[wi] >> ibit) & 1ull) ? 1.0f : 0.0f)
+                            - mlch[t];
                 }
 #pragma unroll
                 for (int tt = 0; tt < GRAM_TJB; ++tt) {
                     if (tt < ntj) {
-                        const int j = (tj0 + tt) * 16 + arow;
                         float bv = 0.0f;
                         if (tv) {
-                            if (j < mm)
-                                bv = w * ((float)((bits[j >> 6] >> (j & 63)) & 1ull) - ml);
-                            else if (j < cols)
-                                bv = w * eych[t * n_out + (j - mm)];
+                            if (jgram[tt])
+                                bv = (((bits[jw[tt]] >> jb[tt]) & 1ull)
+                                          ? w : 0.0f) - wml;
+                            else if (jrhs[tt])
+                                bv = eych[t * n_out + jo[tt]];
                         }
                         acc[tt] = __builtin_amdgcn_mfma_f32_16x16x4f32(
                             a, bv, acc[tt], 0, 0, 0);
@@ -1376,7 +1398,7 @@ extern "C" int launch_wls_gram(
     int NU = 0;   // upper-triangle tile blocks (rhs cols ride the last ones)
     for (int ti = 0; ti < TI; ++ti)
         NU += (TJ - ti + GRAM_TJB - 1) / GRAM_TJB;
-    size_t lds = (size_t)GRAM_CHUNK * W * 8 + GRAM_CHUNK * 4
+    size_t lds = (size_t)GRAM_CHUNK * W * 8 + (size_t)GRAM_CHUNK * 3 * 4
                + (size_t)GRAM_CHUNK * n_out * 4 + (size_t)4 * 16 * 17 * 8;
     wls_gram_kernel<<<dim3(B * NU), dim3(256), lds, stream>>>(
         packed, kw, ey_adj, total, A64, rhs64, B, S, M, W, n_out, NU);
@@ -1475,8 +1497,12 @@ void fused_predict_tiled_kernel(
                         : 0.0f;
             }
             __syncthreads();
-            const int kend = min(KC, Mpad - kc0);
-            for (int ks = 0; ks < kend; ks += 4) {
+            // Mpad is a multiple of KC (launcher contract): the k loop fully
+            // unrolls and every LDS offset is a compile-time constant — the
+            // runtime-bound variant measured 4.9 VALU per MFMA (PMC), all
+            // address math
+#pragma unroll
+            for (int ks = 0; ks < KC; ks += 4) {
                 const int k = kc0 + ks + akcol;
                 const float a =
                     (svalid && k < M) ? (float)(mrow[k] & 1) : 0.0f;
@@ -1639,7 +1665,7 @@ extern "C" int launch_fused_predict_tiled(
     const float* wbg, float* partial, float* ey, int B, int S, int M,
     int Mpad, int Npad, int n_out, int act, hipStream_t stream)
 {
-    if (Npad % 16 != 0 || Mpad % 4 != 0) return -1;
+    if (Npad % 16 != 0 || Mpad % 16 != 0) return -1;  // Mpad % KC == 0
     switch (n_out) {
         case 1: launch_ft_act<1>(masksU, diff, base, wbg, partial, ey, B, S, M, Mpad, Npad, act, stream); break;
         case 2: launch_ft_act<2>(masksU, diff, base, wbg, partial, ey, B, S, M, Mpad, Npad, act, stream); break;
