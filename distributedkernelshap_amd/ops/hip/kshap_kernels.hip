@@ -1486,15 +1486,25 @@ void fused_predict_tiled_kernel(
 
         for (int kc0 = 0; kc0 < Mpad; kc0 += KC) {
             __syncthreads();
-            for (int idx = tid; idx < OIMG * KC * 128; idx += 256) {
-                const int n = idx & 127;
-                const int k = (idx >> 7) % KC;
-                const int o = idx / (128 * KC);
-                const int kg = kc0 + k;
-                diff_lds[(o * KC + k) * FT_NSTRIDE + n] =
-                    (kg < Mpad && n < ncols)
-                        ? dsrc[((size_t)o * Mpad + kg) * Npad + n0 + n]
-                        : 0.0f;
+            // strength-reduced staging: each thread owns one column (and one
+            // k-parity), so the source address just strides by 2*Npad — the
+            // generic idx-decomposition form cost ~15 VALU/element of 64-bit
+            // address math and dominated the kernel's VALU (PMC)
+            {
+                const int sn = tid & 127;
+                const int sk0 = tid >> 7;        // 0 or 1
+                const bool nv = sn < ncols;
+#pragma unroll
+                for (int o = 0; o < OIMG; ++o) {
+                    const float* srcp =
+                        dsrc + ((size_t)o * Mpad + kc0 + sk0) * Npad + n0 + sn;
+                    float* dstp =
+                        diff_lds + (o * KC + sk0) * FT_NSTRIDE + sn;
+#pragma unroll
+                    for (int q = 0; q < KC / 2; ++q)
+                        dstp[2 * q * FT_NSTRIDE] =
+                            nv ? srcp[(size_t)(2 * q) * Npad] : 0.0f;
+                }
             }
             __syncthreads();
             // Mpad is a multiple of KC (launcher contract): the k loop fully
