@@ -1218,6 +1218,7 @@ extern "C" void launch_pack_masks_words(
 #define GRAM_TJB 4
 #define GRAM_MAX_W 8   // M <= 513
 
+template <int W>
 __global__ __launch_bounds__(256)
 void wls_gram_kernel(
     const uint64_t* __restrict__ packed,  // (B, S, W)
@@ -1226,7 +1227,7 @@ void wls_gram_kernel(
     const float* __restrict__ total,      // (B, n_out)
     double* __restrict__ A64,             // (B, mm, mm)
     double* __restrict__ rhs64,           // (B, mm, n_out)
-    int B, int S, int M, int W, int n_out, int NU)
+    int B, int S, int M, int n_out, int NU)
 {
     // one workgroup per (instance, tj-tile-block) work UNIT: a strip-per-ti
     // grid left the chip load-imbalanced (ti=0 owns 4x the tiles of the
@@ -1329,20 +1330,21 @@ void wls_gram_kernel(
                         a = (((bits[wi] >> ibit) & 1ull) ? 1.0f : 0.0f)
                             - mlch[t];
                 }
+                // no ntj guard: tiles past TJ have jgram==jrhs==false, so
+                // they accumulate exact zeros (the runtime predicate kept
+                // the compiler from folding the unrolled body)
 #pragma unroll
                 for (int tt = 0; tt < GRAM_TJB; ++tt) {
-                    if (tt < ntj) {
-                        float bv = 0.0f;
-                        if (tv) {
-                            if (jgram[tt])
-                                bv = (((bits[jw[tt]] >> jb[tt]) & 1ull)
-                                          ? w : 0.0f) - wml;
-                            else if (jrhs[tt])
-                                bv = eych[t * n_out + jo[tt]];
-                        }
-                        acc[tt] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                            a, bv, acc[tt], 0, 0, 0);
+                    float bv = 0.0f;
+                    if (tv) {
+                        if (jgram[tt])
+                            bv = (((bits[jw[tt]] >> jb[tt]) & 1ull)
+                                      ? w : 0.0f) - wml;
+                        else if (jrhs[tt])
+                            bv = eych[t * n_out + jo[tt]];
                     }
+                    acc[tt] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                        a, bv, acc[tt], 0, 0, 0);
                 }
             }
             // promote the chunk partial to fp64 (fp32 run length <= 256)
@@ -1400,8 +1402,19 @@ extern "C" int launch_wls_gram(
         NU += (TJ - ti + GRAM_TJB - 1) / GRAM_TJB;
     size_t lds = (size_t)GRAM_CHUNK * W * 8 + (size_t)GRAM_CHUNK * 3 * 4
                + (size_t)GRAM_CHUNK * n_out * 4 + (size_t)4 * 16 * 17 * 8;
-    wls_gram_kernel<<<dim3(B * NU), dim3(256), lds, stream>>>(
-        packed, kw, ey_adj, total, A64, rhs64, B, S, M, W, n_out, NU);
+    dim3 grid(B * NU), block(256);
+#define KSHAP_GRAM_CASE(WV) \
+    case WV: \
+        wls_gram_kernel<WV><<<grid, block, lds, stream>>>( \
+            packed, kw, ey_adj, total, A64, rhs64, B, S, M, n_out, NU); \
+        break;
+    switch (W) {
+        KSHAP_GRAM_CASE(1) KSHAP_GRAM_CASE(2) KSHAP_GRAM_CASE(3)
+        KSHAP_GRAM_CASE(4) KSHAP_GRAM_CASE(5) KSHAP_GRAM_CASE(6)
+        KSHAP_GRAM_CASE(7) KSHAP_GRAM_CASE(8)
+        default: return -1;
+    }
+#undef KSHAP_GRAM_CASE
     return 0;
 }
 
