@@ -1520,15 +1520,24 @@ void fused_predict_tiled_kernel(
                 }
             }
             __syncthreads();
+            // this lane's A bits for the chunk, loaded in one batch before
+            // the MFMA stream (per-iteration byte loads left ~300-cycle
+            // global latency inside the unrolled loop)
+            uint32_t abits = 0;
+            if (svalid) {
+#pragma unroll
+                for (int q = 0; q < KC / 4; ++q) {
+                    const int k = kc0 + 4 * q + akcol;
+                    abits |= (k < M ? (uint32_t)(mrow[k] & 1) : 0u) << q;
+                }
+            }
             // Mpad is a multiple of KC (launcher contract): the k loop fully
             // unrolls and every LDS offset is a compile-time constant — the
             // runtime-bound variant measured 4.9 VALU per MFMA (PMC), all
             // address math
 #pragma unroll
             for (int ks = 0; ks < KC; ks += 4) {
-                const int k = kc0 + ks + akcol;
-                const float a =
-                    (svalid && k < M) ? (float)(mrow[k] & 1) : 0.0f;
+                const float a = (float)((abits >> (ks >> 2)) & 1u);
 #pragma unroll
                 for (int ct = 0; ct < FT_NTILE; ++ct)
 #pragma unroll
