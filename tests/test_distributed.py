@@ -79,7 +79,9 @@ def _collective_worker(rank, world, port, ret):
     from distributedkernelshap_amd.parallel import explain_sharded, init_distributed
 
     init_distributed(backend="gloo")
-    data = make_adult_like(n_instances=8, n_background=20, seed=2)
+    # 9 instances over 2 ranks: unequal shard counts exercise the
+    # allgather pad/trim branch
+    data = make_adult_like(n_instances=9, n_background=20, seed=2)
     pred = LinearPredictor.random(data.X.shape[1], 2, seed=2)
     eng = KernelShapEngine(
         pred, data.background, groups=data.groups, link="logit", seed=0, device="cpu"
@@ -90,8 +92,17 @@ def _collective_worker(rank, world, port, ret):
     dist.destroy_process_group()
 
 
-def test_collective_gloo_world2(problem):
-    seq = _sequential(problem)
+def test_collective_gloo_world2():
+    # sequential reference on the SAME 9-instance problem the workers build
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+
+    data = make_adult_like(n_instances=9, n_background=20, seed=2)
+    pred = LinearPredictor.random(data.X.shape[1], 2, seed=2)
+    eng = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cpu",
+    )
+    seq = eng.shap_values(data.X)
     ctx = mp.get_context("spawn")
     ret = ctx.Queue()
     port = 29611
@@ -105,7 +116,7 @@ def test_collective_gloo_world2(problem):
     for p in procs:
         p.join(timeout=60)
     for o in range(2):
-        assert np.allclose(sv[o], seq.shap_values[o], rtol=0, atol=1e-10)
+        assert np.allclose(sv[o], seq[o], rtol=0, atol=1e-10)
 
 
 # --------------------------------------------------------------------- #
