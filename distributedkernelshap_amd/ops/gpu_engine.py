@@ -423,6 +423,39 @@ class GpuKernelShap:
             ey[:, lo:hi] = t.einsum("bsno,n->bso", p, wbg64)
         return ey
 
+    def _ey_fused_tiled_bf16(self, masks, X_dev, varying, vidx_t=None):
+        """bf16 matrix-core tiled predict (opt-in predict_dtype for the
+        stress shapes): one v_mfma_f32_16x16x32_bf16 per 32-deep k block —
+        8x the f32 MFMA rate — with the hi+lo split B operand for
+        fp32-grade results ('bf16x2')."""
+        t = self.torch
+        b, s, m = masks.shape
+        mpad = max(32, (m + 31) // 32 * 32)     # FTB_KC contract
+        npad = (self.N + 15) // 16 * 16
+        if vidx_t is None:
+            vidx_t = t.tensor(varying, dtype=t.int64, device=self.device)
+        act, oimg = self._act_oimg()
+        split = 2 if self.engine.kernels.predict_dtype == "bf16x2" else 1
+        diffb = self._buf(
+            f"difftb{m}", (b, split, oimg, npad, mpad), t.bfloat16,
+            zeroed=True,
+        )
+        self.ext.build_diff_bf16_tiled(
+            self._x_part_img(X_dev, act), self._bg_part_img(act), vidx_t,
+            diffb,
+        )
+        base = t.zeros(oimg, npad, device=self.device)
+        base[:, : self.N] = self._base_img(act)
+        wbg = t.zeros(npad, device=self.device)
+        wbg[: self.N] = self.bg_w
+        n_ntiles = (npad + 127) // 128
+        partial = self._buf("ftpart", (b, n_ntiles, s, self.n_out))
+        ey = self._buf("ey", (b, s, self.n_out))
+        self.ext.fused_predict_tiled_bf16(
+            masks, diffb, base, wbg, partial, ey, act
+        )
+        return ey
+
     def _ey_linear_torch(self, masks, X_dev, varying, s_chunk=4096):
         """Library-GEMM fallback for shapes beyond the fused kernel's limits
         (stress configs: M>64 or N>128). ey = reduce(act(mask @ diff + base))."""
@@ -832,11 +865,23 @@ class GpuKernelShap:
                 )
                 if packed is not None:
                     self.ext.pack_masks(masks, packed)
+                split = 2 if kc.predict_dtype == "bf16x2" else 1
+                bf16_tiled_ok = (
+                    kc.predict_dtype in ("bf16", "bf16x2")
+                    and split * (1 if act3 else self.n_out) * 128 * 40 * 2
+                        <= 64 * 1024
+                )
                 if kc.fused_predict and use_bf16:
                     ey = self._ey_fused_bf16(masks, sub_X, varying, packed=None)
-                elif (kc.fused_predict and mpad <= 64 and npad <= 128
+                elif (kc.fused_predict and kc.predict_dtype == "fp32"
+                        and mpad <= 64 and npad <= 128
                         and self.n_out in (1, 2, 4)):
                     ey = self._ey_fused_linear(masks, sub_X, varying)
+                elif (kc.fused_predict and bf16_tiled_ok
+                        and self.n_out in (1, 2, 4)):
+                    # bf16 matrix-core tiled path (opt-in dtype), any shape
+                    ey = self._ey_fused_tiled_bf16(masks, sub_X, varying)
+                    pairwise = act3
                 elif kc.fused_predict and self.n_out in (1, 2, 4):
                     # stress shapes stay on the hand-written MFMA path; the
                     # tiled kernel dual-accumulates (p0, p1) so the pairwise

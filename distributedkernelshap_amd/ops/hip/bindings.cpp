@@ -57,6 +57,16 @@ extern "C" int launch_fused_predict_tiled(
     const float* wbg, float* partial, float* ey, int B, int S, int M,
     int Mpad, int Npad, int n_out, int act, hipStream_t stream);
 
+extern "C" void launch_build_diff_bf16_tiled(
+    const float* xp, const float* bgp, const int64_t* vidx, uint16_t* out,
+    int B, int G, int O, int N, int m, int Mpad, int Npad, int split,
+    hipStream_t stream);
+
+extern "C" int launch_fused_predict_tiled_bf16(
+    const uint8_t* masksU, const uint16_t* diffB, const float* base,
+    const float* wbg, float* partial, float* ey, int B, int S, int M,
+    int Mpad, int Npad, int n_out, int act, int split, hipStream_t stream);
+
 namespace {
 
 #define CHECK_DEV(t) TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be a contiguous device tensor")
@@ -261,6 +271,53 @@ void fused_predict_tiled(
                          "(Mpad%4==0, Npad%16==0, n_out in {1,2,4})");
 }
 
+void build_diff_bf16_tiled(torch::Tensor xp, torch::Tensor bgp,
+                           torch::Tensor vidx, torch::Tensor out) {
+    CHECK_DEV(xp); CHECK_DEV(bgp); CHECK_DEV(vidx); CHECK_DEV(out);
+    TORCH_CHECK(out.dtype() == torch::kBFloat16 && out.dim() == 5,
+                "out must be bf16 (B,split,O,Npad,Mpad)");
+    int B = xp.size(0), G = xp.size(1), O = xp.size(2);
+    int N = bgp.size(0), mcount = vidx.size(0);
+    int split = out.size(1), Npad = out.size(3), Mpad = out.size(4);
+    TORCH_CHECK(out.size(0) == B && out.size(2) == O && N <= Npad
+                && mcount <= Mpad);
+    launch_build_diff_bf16_tiled(
+        xp.data_ptr<float>(), bgp.data_ptr<float>(), vidx.data_ptr<int64_t>(),
+        reinterpret_cast<uint16_t*>(out.data_ptr<at::BFloat16>()), B, G, O, N,
+        mcount, Mpad, Npad, split, current_stream());
+}
+
+void fused_predict_tiled_bf16(
+    torch::Tensor masks, torch::Tensor diffB, torch::Tensor base,
+    torch::Tensor wbg, torch::Tensor partial, torch::Tensor ey, int64_t act) {
+    CHECK_DEV(masks); CHECK_DEV(diffB); CHECK_DEV(base); CHECK_DEV(wbg);
+    CHECK_DEV(partial); CHECK_DEV(ey);
+    TORCH_CHECK(masks.dtype() == torch::kUInt8
+                && diffB.dtype() == torch::kBFloat16);
+    int B = masks.size(0), S = masks.size(1), M = masks.size(2);
+    int split = diffB.size(1), Npad = diffB.size(3), Mpad = diffB.size(4);
+    int n_out = ey.size(2);
+    int oimg = (act == 3) ? 1 : n_out;
+    int n_ntiles = (Npad + 127) / 128;
+    TORCH_CHECK(diffB.size(2) == oimg, "diffB image count vs act");
+    TORCH_CHECK(ey.size(0) == B && ey.size(1) == S, "ey shape");
+    TORCH_CHECK(base.size(0) == oimg && base.size(1) == Npad, "base shape");
+    TORCH_CHECK(wbg.size(0) == Npad, "wbg shape");
+    TORCH_CHECK(partial.size(0) == B && partial.size(1) == n_ntiles
+                && partial.size(2) == S && partial.size(3) == n_out,
+                "partial shape");
+    TORCH_CHECK((size_t)split * oimg * 128 * 40 * 2 <= 64 * 1024,
+                "bf16 tiled LDS footprint (split*oimg too large)");
+    int rc = launch_fused_predict_tiled_bf16(
+        masks.data_ptr<uint8_t>(),
+        reinterpret_cast<const uint16_t*>(diffB.data_ptr<at::BFloat16>()),
+        base.data_ptr<float>(), wbg.data_ptr<float>(),
+        partial.data_ptr<float>(), ey.data_ptr<float>(), B, S, M, Mpad, Npad,
+        n_out, (int)act, split, current_stream());
+    TORCH_CHECK(rc == 0, "fused_predict_tiled_bf16: unsupported shape "
+                         "(Mpad%32==0, Npad%16==0, n_out in {1,2,4})");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -292,4 +349,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_predict_tiled", &fused_predict_tiled,
           "tiled MFMA fused predict for Mpad>64 / Npad>128 (LDS-streamed "
           "diff chunks, per-column-tile partials + deterministic reduce)");
+    m.def("build_diff_bf16_tiled", &build_diff_bf16_tiled,
+          "hi(+lo) k-contiguous diff image for the bf16 tiled predict");
+    m.def("fused_predict_tiled_bf16", &fused_predict_tiled_bf16,
+          "bf16 matrix-core tiled fused predict (v_mfma_f32_16x16x32_bf16 "
+          "per 32-deep k block; hi+lo split for fp32-grade)");
 }

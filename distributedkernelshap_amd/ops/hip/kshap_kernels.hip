@@ -1628,6 +1628,313 @@ void fused_predict_tiled_kernel(
     }
 }
 
+// ------------------------------------------------------------------------- //
+// K3-K6 fused, tiled, bf16 matrix cores (opt-in predict_dtype bf16/bf16x2
+// for the stress shapes): one v_mfma_f32_16x16x32_bf16 per 32-deep k block
+// per column tile — 8x the f32 MFMA rate — with the hi(+lo split) B operand
+// staged per k-block from a k-contiguous global image.  Same grid/partial/
+// reduce structure as the f32 tiled kernel.
+// ------------------------------------------------------------------------- //
+
+#define FTB_KC 32           // k depth of one bf16 MFMA
+
+__global__ void reduce_partials_kernel(
+    const float* __restrict__ partial, float* __restrict__ ey,
+    size_t total_rows, int S, int n_ntiles, int nacc, int n_out);
+
+template <int NOUT, int ACT, int SPLIT>
+__global__ __launch_bounds__(256)
+void fused_predict_tiled_bf16_kernel(
+    const uint8_t* __restrict__ masksU, // (B, S, M)
+    const __bf16* __restrict__ diffB,   // (B, SPLIT, OIMG, Npad, Mpad) k-contig
+    const float* __restrict__ base,     // (OIMG, Npad)
+    const float* __restrict__ wbg,      // (Npad)
+    float* __restrict__ partial_out,    // (B, n_ntiles, S, NACC)
+    int B, int S, int M, int Mpad, int Npad)
+{
+    constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
+    constexpr int NACC = NOUT;
+    const int n_ntiles = (Npad + 127) / 128;
+    const int n_stiles = (S + S_TILE - 1) / S_TILE;
+    const int stile = blockIdx.x % n_stiles;
+    const int nt = (blockIdx.x / n_stiles) % n_ntiles;
+    const int b = blockIdx.x / (n_stiles * n_ntiles);
+    const int n0 = nt * 128;
+    const int ncols = min(128, Npad - n0);
+    const int s0 = stile * S_TILE;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wave = tid >> 6;
+    const int swave = wave * 16;
+    const int arow = lane & 15;          // A row (s) / B col (n)
+    const int akb = lane >> 4;           // k-block 0..3 (8 elements each)
+
+    extern __shared__ float lds[];
+    __bf16* diff_lds = (__bf16*)lds;     // SPLIT*OIMG*128*KSTRIDE_BF
+    float* base_lds = lds + (SPLIT * OIMG * 128 * KSTRIDE_BF + 1) / 2;
+    float* wbg_lds = base_lds + OIMG * 128;
+
+    for (int idx = tid; idx < OIMG * 128; idx += 256) {
+        const int o = idx >> 7, n = idx & 127;
+        base_lds[idx] = (n < ncols) ? base[(size_t)o * Npad + n0 + n] : 0.0f;
+    }
+    for (int idx = tid; idx < 128; idx += 256)
+        wbg_lds[idx] = (idx < ncols) ? wbg[n0 + idx] : 0.0f;
+
+    const __bf16* dsrc = diffB + (size_t)b * SPLIT * OIMG * Npad * Mpad;
+    const uint8_t* mlane = masksU + ((size_t)b * S + swave + arow) * M;
+    const __bf16* dlane = diff_lds + (size_t)arow * KSTRIDE_BF + akb * 8;
+
+    for (int sub = 0; sub < S_TILE / S_SUB; ++sub) {
+        const int ssub0 = s0 + sub * S_SUB;
+        if (ssub0 >= S) break;
+        const int srow = ssub0 + swave + arow;
+        const bool svalid = srow < S;
+        const uint8_t* mrow = mlane + (size_t)ssub0 * M;
+
+        f32x4 acc[FT_NTILE][OIMG];
+#pragma unroll
+        for (int ct = 0; ct < FT_NTILE; ++ct)
+#pragma unroll
+            for (int o = 0; o < OIMG; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
+
+        for (int kc0 = 0; kc0 < Mpad; kc0 += FTB_KC) {
+            __syncthreads();
+            // stage the k-block: one thread per (split*o, n) strip copies 32
+            // contiguous bf16 from the k-contiguous image into the padded
+            // KSTRIDE_BF layout
+            for (int so = tid >> 7; so < SPLIT * OIMG; so += 2) {
+                const int n = tid & 127;
+                const bool nv = n < ncols;
+                const __bf16* sp =
+                    dsrc + ((size_t)so * Npad + n0 + n) * Mpad + kc0;
+                __bf16* dp = diff_lds + ((size_t)so * 128 + n) * KSTRIDE_BF;
+#pragma unroll
+                for (int q = 0; q < FTB_KC / 8; ++q) {
+                    bf16x8 v;
+                    if (nv)
+                        v = *(const bf16x8*)(sp + q * 8);
+                    else
+#pragma unroll
+                        for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
+                    *(bf16x8*)(dp + q * 8) = v;
+                }
+            }
+            __syncthreads();
+            // A fragment: this lane's 8 mask bits for the k-block, converted
+            // in-register (exact in bf16)
+            bf16x8 a;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int k = kc0 + akb * 8 + j;
+                a[j] = (__bf16)(float)(
+                    (svalid && k < M) ? (mrow[k] & 1) : 0);
+            }
+#pragma unroll
+            for (int ct = 0; ct < FT_NTILE; ++ct)
+#pragma unroll
+                for (int o = 0; o < OIMG; ++o)
+#pragma unroll
+                    for (int sp = 0; sp < SPLIT; ++sp) {
+                        const bf16x8 bv = *(const bf16x8*)(
+                            dlane + ((size_t)(sp * OIMG + o) * 128 + ct * 16)
+                                        * KSTRIDE_BF);
+                        acc[ct][o] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a, bv, acc[ct][o], 0, 0, 0);
+                    }
+        }
+
+        float partialv[NACC][4];
+#pragma unroll
+        for (int o = 0; o < NACC; ++o)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) partialv[o][r] = 0.0f;
+#pragma unroll
+        for (int ct = 0; ct < FT_NTILE; ++ct) {
+            const int n = ct * 16 + arow;
+            const float wn = wbg_lds[n];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float z[OIMG];
+#pragma unroll
+                for (int o = 0; o < OIMG; ++o)
+                    z[o] = acc[ct][o][r] + base_lds[o * 128 + n];
+                float zz[NACC];
+                if (ACT == 3) {
+                    const float e = __expf(-z[0]);
+                    const float p1 = fast_rcp(1.0f + e);
+                    zz[1] = p1;
+                    zz[0] = p1 * e;
+                } else if (ACT == 1) {
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o)
+                        zz[o] = fast_rcp(1.0f + __expf(-z[o]));
+                } else if (ACT == 2 && NOUT == 2) {
+                    const float e = __expf(z[0] - z[1]);
+                    const float p1 = fast_rcp(1.0f + e);
+                    zz[0] = p1 * e;
+                    zz[1] = p1;
+                } else if (ACT == 2) {
+                    float mx = z[0];
+#pragma unroll
+                    for (int o = 1; o < NOUT; ++o) mx = fmaxf(mx, z[o]);
+                    float sum = 0.0f;
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) {
+                        zz[o] = __expf(z[o] - mx);
+                        sum += zz[o];
+                    }
+                    const float inv = fast_rcp(sum);
+#pragma unroll
+                    for (int o = 0; o < NOUT; ++o) zz[o] *= inv;
+                } else {
+#pragma unroll
+                    for (int o = 0; o < NACC; ++o) zz[o] = z[o];
+                }
+#pragma unroll
+                for (int o = 0; o < NACC; ++o) partialv[o][r] += wn * zz[o];
+            }
+        }
+#pragma unroll
+        for (int o = 0; o < NACC; ++o)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float v = partialv[o][r];
+                v += __shfl_xor(v, 1);
+                v += __shfl_xor(v, 2);
+                v += __shfl_xor(v, 4);
+                v += __shfl_xor(v, 8);
+                partialv[o][r] = v;
+            }
+        if (arow == 0) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int ss = ssub0 + swave + akb * 4 + r;
+                if (ss < S)
+#pragma unroll
+                    for (int o = 0; o < NACC; ++o)
+                        partial_out[(((size_t)b * n_ntiles + nt) * S + ss)
+                                        * NACC + o] = partialv[o][r];
+            }
+        }
+    }
+}
+
+__global__ void build_diff_bf16_tiled_kernel(
+    const float* __restrict__ xp,    // (B, G, O)
+    const float* __restrict__ bgp,   // (N, G, O)
+    const int64_t* __restrict__ vidx,  // (m,)
+    __bf16* __restrict__ out,        // (B, SPLIT, O, Npad, Mpad) k-contig
+    int G, int O, int N, int m, int Mpad, int Npad, int split, size_t total)
+{
+    size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= total) return;       // total = B*O*N*m, k fastest
+    const int k = idx % m;
+    const int n = (idx / m) % N;
+    const int o = (idx / ((size_t)m * N)) % O;
+    const size_t b = idx / ((size_t)m * N * O);
+    const int64_t g = vidx[k];
+    const float v = xp[((size_t)b * G + g) * O + o]
+                  - bgp[((size_t)n * G + g) * O + o];
+    const __bf16 hi = (__bf16)v;
+    const size_t basei =
+        ((((size_t)b * split) * O + o) * Npad + n) * Mpad + k;
+    out[basei] = hi;
+    if (split == 2) {
+        const size_t lo =
+            ((((size_t)b * split + 1) * O + o) * Npad + n) * Mpad + k;
+        out[lo] = (__bf16)(v - (float)hi);
+    }
+}
+
+extern "C" void launch_build_diff_bf16_tiled(
+    const float* xp, const float* bgp, const int64_t* vidx, uint16_t* out,
+    int B, int G, int O, int N, int m, int Mpad, int Npad, int split,
+    hipStream_t stream)
+{
+    size_t total = (size_t)B * O * N * m;
+    build_diff_bf16_tiled_kernel<<<dim3((unsigned)((total + 255) / 256)),
+                                   dim3(256), 0, stream>>>(
+        xp, bgp, vidx, reinterpret_cast<__bf16*>(out), G, O, N, m, Mpad,
+        Npad, split, total);
+}
+
+template <int NOUT, int ACT, int SPLIT>
+static void launch_ftb_one(
+    const uint8_t* masksU, const __bf16* diffB, const float* base,
+    const float* wbg, float* partial, float* ey, int B, int S, int M,
+    int Mpad, int Npad, hipStream_t stream)
+{
+    constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
+    constexpr int NACC = NOUT;
+    const int n_ntiles = (Npad + 127) / 128;
+    const int n_stiles = (S + S_TILE - 1) / S_TILE;
+    dim3 grid(B * n_ntiles * n_stiles), block(256);
+    size_t lds = ((size_t)SPLIT * OIMG * 128 * KSTRIDE_BF * 2 + 2)
+               + (size_t)(OIMG * 128 + 128) * 4 + 4;
+    fused_predict_tiled_bf16_kernel<NOUT, ACT, SPLIT>
+        <<<grid, block, lds, stream>>>(
+            masksU, diffB, base, wbg, partial, B, S, M, Mpad, Npad);
+    size_t rows = (size_t)B * S;
+    reduce_partials_kernel<<<dim3((unsigned)((rows + 255) / 256)), dim3(256),
+                             0, stream>>>(
+        partial, ey, rows, S, n_ntiles, NACC, NOUT);
+}
+
+template <int NOUT, int ACT>
+static void launch_ftb_split(
+    const uint8_t* masksU, const __bf16* diffB, const float* base,
+    const float* wbg, float* partial, float* ey, int B, int S, int M,
+    int Mpad, int Npad, int split, hipStream_t stream)
+{
+    if (split == 2)
+        launch_ftb_one<NOUT, ACT, 2>(masksU, diffB, base, wbg, partial, ey,
+                                     B, S, M, Mpad, Npad, stream);
+    else
+        launch_ftb_one<NOUT, ACT, 1>(masksU, diffB, base, wbg, partial, ey,
+                                     B, S, M, Mpad, Npad, stream);
+}
+
+template <int NOUT>
+static void launch_ftb_act(
+    const uint8_t* masksU, const __bf16* diffB, const float* base,
+    const float* wbg, float* partial, float* ey, int B, int S, int M,
+    int Mpad, int Npad, int act, int split, hipStream_t stream)
+{
+    switch (act) {
+        case 0:
+            launch_ftb_split<NOUT, 0>(masksU, diffB, base, wbg, partial, ey, B, S, M, Mpad, Npad, split, stream);
+            break;
+        case 1:
+            launch_ftb_split<NOUT, 1>(masksU, diffB, base, wbg, partial, ey, B, S, M, Mpad, Npad, split, stream);
+            break;
+        case 3:
+            if constexpr (NOUT == 2)
+                launch_ftb_split<NOUT, 3>(masksU, diffB, base, wbg, partial, ey, B, S, M, Mpad, Npad, split, stream);
+            break;
+        default:
+            launch_ftb_split<NOUT, 2>(masksU, diffB, base, wbg, partial, ey, B, S, M, Mpad, Npad, split, stream);
+            break;
+    }
+}
+
+extern "C" int launch_fused_predict_tiled_bf16(
+    const uint8_t* masksU, const uint16_t* diffB_u, const float* base,
+    const float* wbg, float* partial, float* ey, int B, int S, int M,
+    int Mpad, int Npad, int n_out, int act, int split, hipStream_t stream)
+{
+    if (Npad % 16 != 0 || Mpad % FTB_KC != 0 || split < 1 || split > 2)
+        return -1;
+    const __bf16* diffB = reinterpret_cast<const __bf16*>(diffB_u);
+    switch (n_out) {
+        case 1: launch_ftb_act<1>(masksU, diffB, base, wbg, partial, ey, B, S, M, Mpad, Npad, act, split, stream); break;
+        case 2: launch_ftb_act<2>(masksU, diffB, base, wbg, partial, ey, B, S, M, Mpad, Npad, act, split, stream); break;
+        case 4: launch_ftb_act<4>(masksU, diffB, base, wbg, partial, ey, B, S, M, Mpad, Npad, act, split, stream); break;
+        default: return -1;
+    }
+    return 0;
+}
+
 // deterministic column-tile reduction: ey[b,s,:] from the per-tile partials
 // (a fixed summation order — float atomics would break the bitwise-
 // determinism guarantee the GPU tests assert)

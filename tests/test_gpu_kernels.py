@@ -868,3 +868,59 @@ def test_engine_gpu_fp64_mode_matches_oracle():
     sv_f = g32.shap_values(X)
     errs = [np.abs(sv_f[o] - sv_g[o]).max() for o in range(2)]
     assert max(errs) < 1e-3
+
+
+def test_fused_predict_tiled_bf16_vs_fp32():
+    """bf16x2 tiled predict (v_mfma_f32_16x16x32_bf16, hi+lo split) matches
+    the f32 tiled kernel to fp32-grade tolerance on stress shapes."""
+    from distributedkernelshap_amd.config import KernelConfig
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.models import LinearPredictor, make_tabular
+
+    for m, n_bg, act, n_out in [(200, 300, "softmax", 2), (96, 200, "sigmoid", 2)]:
+        data = make_tabular(n_features=m, n_instances=3, n_background=n_bg,
+                            seed=4)
+        pred = LinearPredictor.random(m, n_out, seed=4, activation=act)
+        eng = KernelShapEngine(
+            pred, data.background, groups=data.groups, link="identity",
+            seed=0, device="cuda",
+            kernels=KernelConfig(predict_dtype="bf16x2"),
+        )
+        gpu = eng._gpu
+        varying = np.arange(m)
+        plan = eng._plan(m, 1024)
+        masks, _kw = gpu._device_masks(plan, np.arange(3))
+        X_dev = torch.tensor(data.X, dtype=torch.float32, device="cuda")
+        ey_b = gpu._ey_fused_tiled_bf16(masks, X_dev, varying).clone()
+        ey_f = gpu._ey_fused_tiled(masks, X_dev, varying).clone()
+        err = (ey_b - ey_f).abs().max().item()
+        assert err < 2e-3, (m, n_bg, act, err)
+
+
+def test_engine_gpu_stress_bf16x2_local_accuracy():
+    """End-to-end stress shape on the bf16x2 tiled path: local accuracy and
+    fp32-grade agreement with the fp32 engine."""
+    from distributedkernelshap_amd.config import KernelConfig
+    from distributedkernelshap_amd.core.engine import KernelShapEngine
+    from distributedkernelshap_amd.core.links import logit
+    from distributedkernelshap_amd.models import LinearPredictor, make_tabular
+
+    data = make_tabular(n_features=200, n_instances=4, n_background=150, seed=1)
+    pred = LinearPredictor.random(200, 2, seed=1)
+    engb = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cuda", kernels=KernelConfig(predict_dtype="bf16x2"),
+    )
+    svb = engb.shap_values(data.X, nsamples=2048, l1_reg=False)
+    fx = logit(pred(data.X))
+    for o in range(2):
+        total = svb[o].sum(axis=1) + engb.expected_value[o]
+        assert np.abs(total - fx[:, o]).max() < 2e-3
+    engf = KernelShapEngine(
+        pred, data.background, groups=data.groups, link="logit", seed=0,
+        device="cuda",
+    )
+    svf = engf.shap_values(data.X, nsamples=2048, l1_reg=False)
+    for o in range(2):
+        err = np.abs(svb[o] - svf[o]).max()
+        assert err < 5e-2, err
