@@ -211,3 +211,13 @@ def sample_masks(
         weights[ne:] = plan.weight_left / n_rand
 
     return masks, weights
+
+
+def sampler_chunks(n_random: int) -> int:
+    """Chunk count of the GPU sampler's deterministic per-wave partitioning
+    (mirrors ``kshap_sampler_chunks`` in kshap_kernels.hip): ~256 rows per
+    chunk, a multiple of 8 in [8, 64]. Plans with n_random <= 2048 keep the
+    historic 8 chunks, so their device masks are bit-for-bit stable."""
+    c = (n_random + 255) // 256
+    c = (c + 7) // 8 * 8
+    return max(8, min(64, c))

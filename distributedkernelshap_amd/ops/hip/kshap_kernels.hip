@@ -111,16 +111,21 @@ __global__ void fill_random_masks_kernel(
     int n_sizes,
     int num_paired,                   // sizes <= num_paired get a complement
     uint32_t seed,
-    const int32_t* __restrict__ inst_ids)  // (B,) global instance index (RNG key)
+    const int32_t* __restrict__ inst_ids,  // (B,) global instance index (RNG key)
+    int n_chunks)                     // deterministic chunk count (multiple of 8)
 {
-    const int b = blockIdx.x;
+    const int nblk = n_chunks / 8;    // blocks per instance, 8 waves each
+    const int b = blockIdx.x / nblk;
+    const int cb = blockIdx.x % nblk;
     if (b >= B) return;
     const int lane = threadIdx.x & (WAVE - 1);
-    const int wv = threadIdx.x >> 6;     // 8 waves split the row range
-    // fixed per-wave chunking keeps the output deterministic regardless of
-    // launch timing; complement pairs never cross a chunk boundary
-    const int chunk = (n_random + 7) / 8;
-    const int lo = wv * chunk;
+    const int chunk_id = cb * 8 + (threadIdx.x >> 6);
+    // fixed chunking (a deterministic function of n_random only) keeps the
+    // output reproducible regardless of launch timing; complement pairs
+    // never cross a chunk boundary.  n_chunks scales with n_random so big
+    // plans (stress: 16k rows) fill the chip instead of B blocks
+    const int chunk = (n_random + n_chunks - 1) / n_chunks;
+    const int lo = chunk_id * chunk;
     const int hi = min(lo + chunk, n_random);
     if (lo >= hi) return;
 
@@ -134,7 +139,7 @@ __global__ void fill_random_masks_kernel(
     while (remaining > 0) {
         Philox rng;
         rng.init(seed, (uint32_t)inst_ids[b], iter,
-                 (uint32_t)(wv * WAVE + lane) | 0x52000000u);
+                 (uint32_t)(chunk_id * WAVE + lane) | 0x52000000u);
         // draw subset size from the residual kernel distribution
         float u = (rng.next_u32() >> 8) * (1.0f / 16777216.0f);
         int si = 0;
@@ -200,6 +205,17 @@ __global__ void fill_random_masks_kernel(
     }
 }
 
+extern "C" int kshap_sampler_chunks(int n_random)
+{
+    // deterministic chunk count: ~256 rows per chunk, multiple of 8 in
+    // [8, 64] (n_random <= 2048 keeps the historic 8 chunks bit-for-bit)
+    int c = (n_random + 255) / 256;
+    c = (c + 7) / 8 * 8;
+    if (c < 8) c = 8;
+    if (c > 64) c = 64;
+    return c;
+}
+
 extern "C" void launch_fill_random_masks(
     uint8_t* masks, int B, int S, int M, int ne, int n_random,
     const float* cdf, const int* sizes, int n_sizes, int num_paired,
@@ -207,26 +223,28 @@ extern "C" void launch_fill_random_masks(
 {
     if (n_random <= 0 || B <= 0) return;
     const int words = (M + 63) / 64;
+    const int n_chunks = kshap_sampler_chunks(n_random);
+    dim3 grid(B * (n_chunks / 8)), block(8 * WAVE);
     switch (words) {
         case 1:
-            fill_random_masks_kernel<1><<<dim3(B), dim3(8 * WAVE), 0, stream>>>(
+            fill_random_masks_kernel<1><<<grid, block, 0, stream>>>(
                 masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
-                seed, inst_ids);
+                seed, inst_ids, n_chunks);
             break;
         case 2:
-            fill_random_masks_kernel<2><<<dim3(B), dim3(8 * WAVE), 0, stream>>>(
+            fill_random_masks_kernel<2><<<grid, block, 0, stream>>>(
                 masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
-                seed, inst_ids);
+                seed, inst_ids, n_chunks);
             break;
         case 3:
-            fill_random_masks_kernel<3><<<dim3(B), dim3(8 * WAVE), 0, stream>>>(
+            fill_random_masks_kernel<3><<<grid, block, 0, stream>>>(
                 masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
-                seed, inst_ids);
+                seed, inst_ids, n_chunks);
             break;
         default:
-            fill_random_masks_kernel<4><<<dim3(B), dim3(8 * WAVE), 0, stream>>>(
+            fill_random_masks_kernel<4><<<grid, block, 0, stream>>>(
                 masks, B, S, M, ne, n_random, cdf, sizes, n_sizes, num_paired,
-                seed, inst_ids);
+                seed, inst_ids, n_chunks);
             break;
     }
 }
@@ -1215,7 +1233,7 @@ extern "C" void launch_pack_masks_words(
 // ------------------------------------------------------------------------- //
 
 #define GRAM_CHUNK 256
-#define GRAM_TJB 4
+#define GRAM_TJB 8
 #define GRAM_MAX_W 8   // M <= 513
 
 template <int W>

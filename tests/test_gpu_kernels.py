@@ -43,10 +43,13 @@ def test_fill_random_masks_properties(ext):
     sizes = rnd.sum(axis=2)
     # all rows filled with sizes from the residual range (draw 4-6, compl 6-8)
     assert sizes.min() >= 4 and sizes.max() <= 8
-    # complement pairing within each of the 8 fixed per-wave chunks (a pair
-    # never crosses a chunk boundary; the last row of a chunk may be an
-    # unpaired truncated draw)
-    chunk = (plan.n_random + 7) // 8
+    # complement pairing within each fixed per-wave chunk (a pair never
+    # crosses a chunk boundary; the last row of a chunk may be an unpaired
+    # truncated draw)
+    from distributedkernelshap_amd.core.sampler import sampler_chunks
+
+    chunk = (plan.n_random + sampler_chunks(plan.n_random) - 1) \
+        // sampler_chunks(plan.n_random)
     for bi in range(b):
         for clo in range(0, plan.n_random, chunk):
             chi = min(clo + chunk, plan.n_random)
@@ -126,7 +129,10 @@ def test_fill_random_masks_size_distribution(ext):
         exp[s] /= zsum
     # every emitted size must be an expected one
     assert set(np.unique(sizes)).issubset(set(exp))
-    slack = 8 * b  # ≤1 truncated complement per fixed per-wave chunk
+    from distributedkernelshap_amd.core.sampler import sampler_chunks
+
+    # ≤1 truncated complement per fixed per-wave chunk
+    slack = sampler_chunks(plan.n_random) * b
     for s, p in exp.items():
         cnt = int((sizes == s).sum())
         sigma = (n * p * (1 - p)) ** 0.5
@@ -529,7 +535,10 @@ def test_fill_random_masks_wide_m(ext):
     lo, hi = plan.random_sizes.min(), plan.random_sizes.max()
     assert sizes.min() >= lo and sizes.max() <= m - lo
     # paired draws followed by exact complements (within per-wave chunks)
-    chunk = (plan.n_random + 7) // 8
+    from distributedkernelshap_amd.core.sampler import sampler_chunks
+
+    chunk = (plan.n_random + sampler_chunks(plan.n_random) - 1) \
+        // sampler_chunks(plan.n_random)
     for bi in range(2):
         for clo in range(0, plan.n_random, chunk):
             chi = min(clo + chunk, plan.n_random)
