@@ -1233,7 +1233,7 @@ extern "C" void launch_pack_masks_words(
 // ------------------------------------------------------------------------- //
 
 #define GRAM_CHUNK 256
-#define GRAM_TJB 8
+#define GRAM_TJB 4
 #define GRAM_MAX_W 8   // M <= 513
 
 template <int W>
