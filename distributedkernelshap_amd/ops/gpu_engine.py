@@ -672,7 +672,10 @@ class GpuKernelShap:
             # configs: the whole-batch mask tensor would be TB-scale)
             if as_tensor:
                 out = t.empty(
-                    b_total, self.n_groups, self.n_out, device=self.device
+                    b_total, self.n_groups, self.n_out, device=self.device,
+                    dtype=(t.float64
+                           if self.engine.kernels.predict_dtype == "fp64"
+                           else t.float32),
                 )
                 for lo in range(0, b_total, chunk):
                     hi = min(lo + chunk, b_total)

@@ -51,10 +51,9 @@ lfx = gpu._link(fx64)
 lfnull64 = gpu._link(gpu.fnull.double())
 total_all = (lfx - lfnull64[None]).float()
 lfnull = lfnull64.float()
-eyc = ey_t.clone()
-eyc.clamp_(1e-7, 1 - 1e-7)
-eyc.log_().sub_(t.log1p(-t.exp(eyc)))
-ey_adj = eyc.sub_(lfnull[None, None, :])
+# the engine's saturation-safe pairwise link (the clamp-based transform
+# differs by O(1) on saturated rows — that was the round-2 accuracy bug)
+ey_adj = gpu._link_ey(ey_t.clone(), lfnull, True)
 totalc = total_all.contiguous()
 phi_gram = gpu._solve_gram(masks, kw, ey_adj, totalc).clone()
 phi_torch = gpu._solve_torch(masks, kw, ey_adj, totalc)
@@ -81,9 +80,10 @@ print("gram r err:", (r64 - r_ref).abs().max().item(),
       "scale", r_ref.abs().max().item())
 print("cond(A):", np.linalg.cond(a_ref[0].cpu().numpy()))
 print("kw range:", kw.min().item(), kw.max().item())
-# oracle phi via fp64 solve on fp64 ey for isolation
-eyc64 = ey_64.clamp(1e-15, 1 - 1e-15)
-ey_adj64 = t.log(eyc64 / (1 - eyc64)) - lfnull64[None, None, :]
+# oracle phi via fp64 solve on fp64 ey (same pairwise link) for isolation
+lr = (t.log(ey_64[..., 1].clamp_min(1e-300))
+      - t.log(ey_64[..., 0].clamp_min(1e-300)))
+ey_adj64 = t.stack([-lr, lr], dim=-1) - lfnull64[None, None, :]
 phi_64 = gpu._solve_torch(masks, kw, ey_adj64.float(), totalc,
                           out_dtype=t.float64)
 print("gram phi vs fp64-ey phi:", (phi_gram.double() - phi_64).abs().max().item())
