@@ -76,7 +76,6 @@ def batched_lars_select(
     kmax = m if num_features is None else min(num_features, m)
     if max_steps is None:
         max_steps = 8 * m if lasso else kmax
-    bidx = t.arange(b, device=dev)
 
     K = t.zeros(b, m, m, dtype=t.float64, device=dev)   # inv(G_AA), slot order
     act = t.full((b, m), m, dtype=t.int64, device=dev)  # slot -> feature (m = pad)
@@ -190,7 +189,7 @@ def batched_lars_select(
             sgn[ai, k[ai]] = t.sign(rj[ai])
             in_act[ai, jstar[ai]] = True
             k = k + app.long()
-            kcap = min(int(k.max().item()) + 1, m)
+            kcap = min(_step + 2, m)   # still sync-free: k <= _step+1
 
         # equiangular direction over the (updated) active sets of live rows
         actc = act[:, :kcap]
