@@ -1468,9 +1468,10 @@ void fused_predict_tiled_kernel(
     // their logits; the reference's fp64 numpy has no such cliff)
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
     constexpr int NACC = NOUT;
-    // 16-deep k chunks keep the LDS footprint ~10 KB -> 6 workgroups/CU
-    // (32-deep measured slower: 19 KB capped occupancy at 3/CU)
-    constexpr int KC = 16;                      // LDS diff chunk k-depth
+    // 16-deep k chunks keep the double-buffered LDS footprint ~19 KB
+    constexpr int KC = (OIMG >= 4) ? 8 : 16;    // LDS diff chunk k-depth
+    constexpr int BUFSZ = OIMG * KC * FT_NSTRIDE;
+    constexpr int SREG = OIMG * KC / 2;         // staged elems per thread
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     const int stile = blockIdx.x % n_stiles;
@@ -1487,9 +1488,9 @@ void fused_predict_tiled_kernel(
     const int akcol = lane >> 4;
 
     extern __shared__ float lds[];
-    float* diff_lds = lds;                          // OIMG * KC * FT_NSTRIDE
-    float* base_lds = diff_lds + OIMG * KC * FT_NSTRIDE;  // OIMG * 128
-    float* wbg_lds = base_lds + OIMG * 128;               // 128
+    float* diff_lds = lds;                          // 2 * BUFSZ (ping-pong)
+    float* base_lds = diff_lds + 2 * BUFSZ;         // OIMG * 128
+    float* wbg_lds = base_lds + OIMG * 128;         // 128
 
     for (int idx = tid; idx < OIMG * 128; idx += 256) {
         const int o = idx >> 7, n = idx & 127;
@@ -1515,32 +1516,52 @@ void fused_predict_tiled_kernel(
 #pragma unroll
             for (int o = 0; o < OIMG; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
 
-        for (int kc0 = 0; kc0 < Mpad; kc0 += KC) {
-            __syncthreads();
-            // strength-reduced staging: each thread owns one column (and one
-            // k-parity), so the source address just strides by 2*Npad — the
-            // generic idx-decomposition form cost ~15 VALU/element of 64-bit
-            // address math and dominated the kernel's VALU (PMC)
-            {
-                const int sn = tid & 127;
-                const int sk0 = tid >> 7;        // 0 or 1
-                const bool nv = sn < ncols;
+        // software-pipelined ping-pong staging: chunk i+1's global loads are
+        // issued (into registers) while chunk i's MFMAs run from the other
+        // LDS buffer, so the VMEM latency hides and there is ONE barrier per
+        // chunk. Staging is strength-reduced: each thread owns one column
+        // and one k-parity, addresses stride by 2*Npad (the generic
+        // idx-decomposition form cost ~15 VALU/element, PMC).
+        const int sn = tid & 127;
+        const int sk0 = tid >> 7;            // 0 or 1
+        const bool nv = sn < ncols;
+        const int nchunks = Mpad / KC;
+        float sreg[SREG];
+        // preload + write chunk 0 into buffer 0
+#pragma unroll
+        for (int o = 0; o < OIMG; ++o) {
+            const float* srcp = dsrc + ((size_t)o * Mpad + sk0) * Npad + n0 + sn;
+#pragma unroll
+            for (int q = 0; q < KC / 2; ++q)
+                sreg[o * (KC / 2) + q] =
+                    nv ? srcp[(size_t)(2 * q) * Npad] : 0.0f;
+        }
+        __syncthreads();                    // previous sub's reads done
+#pragma unroll
+        for (int o = 0; o < OIMG; ++o) {
+            float* dstp = diff_lds + (o * KC + sk0) * FT_NSTRIDE + sn;
+#pragma unroll
+            for (int q = 0; q < KC / 2; ++q)
+                dstp[2 * q * FT_NSTRIDE] = sreg[o * (KC / 2) + q];
+        }
+
+        for (int kci = 0; kci < nchunks; ++kci) {
+            const int kc0 = kci * KC;
+            const int cur = kci & 1;
+            __syncthreads();                 // buffer `cur` writes visible
+            if (kci + 1 < nchunks) {
+                // issue next chunk's global loads; latency overlaps the MFMAs
 #pragma unroll
                 for (int o = 0; o < OIMG; ++o) {
-                    const float* srcp =
-                        dsrc + ((size_t)o * Mpad + kc0 + sk0) * Npad + n0 + sn;
-                    float* dstp =
-                        diff_lds + (o * KC + sk0) * FT_NSTRIDE + sn;
+                    const float* srcp = dsrc
+                        + ((size_t)o * Mpad + kc0 + KC + sk0) * Npad + n0 + sn;
 #pragma unroll
                     for (int q = 0; q < KC / 2; ++q)
-                        dstp[2 * q * FT_NSTRIDE] =
+                        sreg[o * (KC / 2) + q] =
                             nv ? srcp[(size_t)(2 * q) * Npad] : 0.0f;
                 }
             }
-            __syncthreads();
-            // this lane's A bits for the chunk, loaded in one batch before
-            // the MFMA stream (per-iteration byte loads left ~300-cycle
-            // global latency inside the unrolled loop)
+            // this lane's A bits for the chunk, one batch before the MFMAs
             uint32_t abits = 0;
             if (svalid) {
 #pragma unroll
@@ -1549,10 +1570,9 @@ void fused_predict_tiled_kernel(
                     abits |= (k < M ? (uint32_t)(mrow[k] & 1) : 0u) << q;
                 }
             }
+            const float* dl = dlane + cur * BUFSZ;
             // Mpad is a multiple of KC (launcher contract): the k loop fully
-            // unrolls and every LDS offset is a compile-time constant — the
-            // runtime-bound variant measured 4.9 VALU per MFMA (PMC), all
-            // address math
+            // unrolls and every LDS offset is a compile-time constant
 #pragma unroll
             for (int ks = 0; ks < KC; ks += 4) {
                 const float a = (float)((abits >> (ks >> 2)) & 1u);
@@ -1561,10 +1581,23 @@ void fused_predict_tiled_kernel(
 #pragma unroll
                     for (int o = 0; o < OIMG; ++o) {
                         const float bv =
-                            dlane[(o * KC + ks) * FT_NSTRIDE + ct * 16];
+                            dl[(o * KC + ks) * FT_NSTRIDE + ct * 16];
                         acc[ct][o] = __builtin_amdgcn_mfma_f32_16x16x4f32(
                             a, bv, acc[ct][o], 0, 0, 0);
                     }
+            }
+            if (kci + 1 < nchunks) {
+                // write next chunk into the other buffer; no barrier needed
+                // before the write (everyone passed this iteration's barrier,
+                // so no one still reads that buffer)
+                float* dst0 = diff_lds + (1 - cur) * BUFSZ;
+#pragma unroll
+                for (int o = 0; o < OIMG; ++o) {
+                    float* dstp = dst0 + (o * KC + sk0) * FT_NSTRIDE + sn;
+#pragma unroll
+                    for (int q = 0; q < KC / 2; ++q)
+                        dstp[2 * q * FT_NSTRIDE] = sreg[o * (KC / 2) + q];
+                }
             }
         }
 
@@ -1981,11 +2014,11 @@ static void launch_ft(
 {
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
     constexpr int NACC = NOUT;
-    constexpr int KC = 16;
+    constexpr int KC = (OIMG >= 4) ? 8 : 16;
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     dim3 grid(B * n_ntiles * n_stiles), block(256);
-    size_t lds = (size_t)(OIMG * KC * FT_NSTRIDE + OIMG * 128 + 128) * 4;
+    size_t lds = (size_t)(2 * OIMG * KC * FT_NSTRIDE + OIMG * 128 + 128) * 4;
     fused_predict_tiled_kernel<NOUT, ACT><<<grid, block, lds, stream>>>(
         masksU, diff, base, wbg, partial, B, S, M, Mpad, Npad);
     size_t rows = (size_t)B * S;
