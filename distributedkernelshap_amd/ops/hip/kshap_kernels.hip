@@ -35,6 +35,25 @@
 //                                 back-substitution of the eliminated
 //                                 feature.
 //
+// Stress-shape set (M > 64 or background > 128; round 2):
+//   K2c pack_masks_words        — wide-M packing, one thread per output
+//                                 word (coalesced 64B source spans)
+//   K3-K6 fused_predict_tiled / fused_predict_tiled_bf16 —
+//                                 column-tiled fused predict: diff streamed
+//                                 through double-buffered LDS k-chunks
+//                                 (software-pipelined: next chunk's global
+//                                 loads issue during this chunk's MFMAs),
+//                                 per-column-tile partial reductions summed
+//                                 by a deterministic reduce kernel; bf16
+//                                 variant runs one v_mfma_f32_16x16x32_bf16
+//                                 per 32-deep k block with a hi+lo split B
+//   K7b wls_gram                — (M-1) x (M-1+n_out) normal equations in
+//                                 16x16 MFMA tiles over packed bits, fp64
+//                                 via 256-sample chunked promotion; also
+//                                 powers the batched L1 path (phantom last
+//                                 feature) — the solve itself is library
+//                                 fp64 (S-independent)
+//
 // All kernels are wave64 / LDS-tiled for CDNA4; fp32 compute by default,
 // bf16 matrix-core modes opt-in (engine KernelConfig.predict_dtype).
 
