@@ -1724,6 +1724,11 @@ void fused_predict_tiled_bf16_kernel(
 {
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
     constexpr int NACC = NOUT;
+    constexpr int BUFSZB = SPLIT * OIMG * 128 * KSTRIDE_BF;  // bf16 elements
+    // double-buffered (software-pipelined) staging when two buffers fit the
+    // 64 KB workgroup LDS; larger image sets keep the two-barrier flow
+    constexpr bool DBUF = (SPLIT * OIMG) <= 2;
+    constexpr int NBUF = DBUF ? 2 : 1;
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     const int stile = blockIdx.x % n_stiles;
@@ -1740,8 +1745,8 @@ void fused_predict_tiled_bf16_kernel(
     const int akb = lane >> 4;           // k-block 0..3 (8 elements each)
 
     extern __shared__ float lds[];
-    __bf16* diff_lds = (__bf16*)lds;     // SPLIT*OIMG*128*KSTRIDE_BF
-    float* base_lds = lds + (SPLIT * OIMG * 128 * KSTRIDE_BF + 1) / 2;
+    __bf16* diff_lds = (__bf16*)lds;     // NBUF * BUFSZB
+    float* base_lds = lds + (NBUF * BUFSZB + 1) / 2;
     float* wbg_lds = base_lds + OIMG * 128;
 
     for (int idx = tid; idx < OIMG * 128; idx += 256) {
@@ -1768,29 +1773,79 @@ void fused_predict_tiled_bf16_kernel(
 #pragma unroll
             for (int o = 0; o < OIMG; ++o) acc[ct][o] = (f32x4){0, 0, 0, 0};
 
-        for (int kc0 = 0; kc0 < Mpad; kc0 += FTB_KC) {
-            __syncthreads();
-            // stage the k-block: one thread per (split*o, n) strip copies 32
-            // contiguous bf16 from the k-contiguous image into the padded
-            // KSTRIDE_BF layout
-            for (int so = tid >> 7; so < SPLIT * OIMG; so += 2) {
-                const int n = tid & 127;
-                const bool nv = n < ncols;
+        // stage the k-block: one thread per (split*o, n) strip copies 32
+        // contiguous bf16 from the k-contiguous image into the padded
+        // KSTRIDE_BF layout.  With DBUF, chunk i+1's global loads are issued
+        // into registers during chunk i's MFMAs (one barrier per chunk).
+        const int sso = tid >> 7;
+        const int ssn = tid & 127;
+        const bool snv = ssn < ncols;
+        const int nchunks = Mpad / FTB_KC;
+        bf16x8 sreg[FTB_KC / 8];
+        if constexpr (DBUF) {
+            if (sso < SPLIT * OIMG) {
                 const __bf16* sp =
-                    dsrc + ((size_t)so * Npad + n0 + n) * Mpad + kc0;
-                __bf16* dp = diff_lds + ((size_t)so * 128 + n) * KSTRIDE_BF;
+                    dsrc + ((size_t)sso * Npad + n0 + ssn) * Mpad;
 #pragma unroll
                 for (int q = 0; q < FTB_KC / 8; ++q) {
-                    bf16x8 v;
-                    if (nv)
-                        v = *(const bf16x8*)(sp + q * 8);
+                    if (snv)
+                        sreg[q] = *(const bf16x8*)(sp + q * 8);
                     else
 #pragma unroll
-                        for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
-                    *(bf16x8*)(dp + q * 8) = v;
+                        for (int j = 0; j < 8; ++j) sreg[q][j] = (__bf16)0.0f;
                 }
             }
-            __syncthreads();
+        }
+        __syncthreads();                 // prior sub's reads complete
+        if constexpr (DBUF) {
+            if (sso < SPLIT * OIMG) {
+                __bf16* dp =
+                    diff_lds + ((size_t)sso * 128 + ssn) * KSTRIDE_BF;
+#pragma unroll
+                for (int q = 0; q < FTB_KC / 8; ++q)
+                    *(bf16x8*)(dp + q * 8) = sreg[q];
+            }
+        }
+
+        for (int kci = 0; kci < nchunks; ++kci) {
+            const int kc0 = kci * FTB_KC;
+            const int cur = DBUF ? (kci & 1) : 0;
+            if constexpr (DBUF) {
+                __syncthreads();         // buffer `cur` writes visible
+                if (kci + 1 < nchunks && sso < SPLIT * OIMG) {
+                    const __bf16* sp = dsrc
+                        + ((size_t)sso * Npad + n0 + ssn) * Mpad + kc0 + FTB_KC;
+#pragma unroll
+                    for (int q = 0; q < FTB_KC / 8; ++q) {
+                        if (snv)
+                            sreg[q] = *(const bf16x8*)(sp + q * 8);
+                        else
+#pragma unroll
+                            for (int j = 0; j < 8; ++j)
+                                sreg[q][j] = (__bf16)0.0f;
+                    }
+                }
+            } else {
+                __syncthreads();
+                for (int so = sso; so < SPLIT * OIMG; so += 2) {
+                    const __bf16* sp =
+                        dsrc + ((size_t)so * Npad + n0 + ssn) * Mpad + kc0;
+                    __bf16* dp =
+                        diff_lds + ((size_t)so * 128 + ssn) * KSTRIDE_BF;
+#pragma unroll
+                    for (int q = 0; q < FTB_KC / 8; ++q) {
+                        bf16x8 v;
+                        if (snv)
+                            v = *(const bf16x8*)(sp + q * 8);
+                        else
+#pragma unroll
+                            for (int j = 0; j < 8; ++j)
+                                v[j] = (__bf16)0.0f;
+                        *(bf16x8*)(dp + q * 8) = v;
+                    }
+                }
+                __syncthreads();
+            }
             // A fragment: this lane's 8 mask bits for the k-block, converted
             // in-register (exact in bf16)
             bf16x8 a;
@@ -1800,6 +1855,7 @@ void fused_predict_tiled_bf16_kernel(
                 a[j] = (__bf16)(float)(
                     (svalid && k < M) ? (mrow[k] & 1) : 0);
             }
+            const __bf16* dl = dlane + (size_t)cur * BUFSZB;
 #pragma unroll
             for (int ct = 0; ct < FT_NTILE; ++ct)
 #pragma unroll
@@ -1807,11 +1863,20 @@ void fused_predict_tiled_bf16_kernel(
 #pragma unroll
                     for (int sp = 0; sp < SPLIT; ++sp) {
                         const bf16x8 bv = *(const bf16x8*)(
-                            dlane + ((size_t)(sp * OIMG + o) * 128 + ct * 16)
-                                        * KSTRIDE_BF);
+                            dl + ((size_t)(sp * OIMG + o) * 128 + ct * 16)
+                                     * KSTRIDE_BF);
                         acc[ct][o] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                             a, bv, acc[ct][o], 0, 0, 0);
                     }
+            if constexpr (DBUF) {
+                if (kci + 1 < nchunks && sso < SPLIT * OIMG) {
+                    __bf16* dp = diff_lds + (size_t)(1 - cur) * BUFSZB
+                                 + ((size_t)sso * 128 + ssn) * KSTRIDE_BF;
+#pragma unroll
+                    for (int q = 0; q < FTB_KC / 8; ++q)
+                        *(bf16x8*)(dp + q * 8) = sreg[q];
+                }
+            }
         }
 
         float partialv[NACC][4];
@@ -1937,10 +2002,11 @@ static void launch_ftb_one(
 {
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
     constexpr int NACC = NOUT;
+    constexpr int NBUF = ((SPLIT * OIMG) <= 2) ? 2 : 1;  // mirror DBUF
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     dim3 grid(B * n_ntiles * n_stiles), block(256);
-    size_t lds = ((size_t)SPLIT * OIMG * 128 * KSTRIDE_BF * 2 + 2)
+    size_t lds = ((size_t)NBUF * SPLIT * OIMG * 128 * KSTRIDE_BF * 2 + 2)
                + (size_t)(OIMG * 128 + 128) * 4 + 4;
     fused_predict_tiled_bf16_kernel<NOUT, ACT, SPLIT>
         <<<grid, block, lds, stream>>>(
