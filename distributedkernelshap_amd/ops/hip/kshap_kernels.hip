@@ -1725,9 +1725,10 @@ void fused_predict_tiled_bf16_kernel(
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
     constexpr int NACC = NOUT;
     constexpr int BUFSZB = SPLIT * OIMG * 128 * KSTRIDE_BF;  // bf16 elements
-    // double-buffered (software-pipelined) staging when two buffers fit the
-    // 64 KB workgroup LDS; larger image sets keep the two-barrier flow
-    constexpr bool DBUF = (SPLIT * OIMG) <= 2;
+    // double-buffered (software-pipelined) staging only for the single-
+    // image case (~10 KB/buffer): at SPLIT*OIMG==2 the 41 KB pair capped
+    // occupancy at 1 workgroup/CU and measured slower
+    constexpr bool DBUF = (SPLIT * OIMG) == 1;
     constexpr int NBUF = DBUF ? 2 : 1;
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
@@ -2002,7 +2003,7 @@ static void launch_ftb_one(
 {
     constexpr int OIMG = (ACT == 3) ? 1 : NOUT;
     constexpr int NACC = NOUT;
-    constexpr int NBUF = ((SPLIT * OIMG) <= 2) ? 2 : 1;  // mirror DBUF
+    constexpr int NBUF = ((SPLIT * OIMG) == 1) ? 2 : 1;  // mirror DBUF
     const int n_ntiles = (Npad + 127) / 128;
     const int n_stiles = (S + S_TILE - 1) / S_TILE;
     dim3 grid(B * n_ntiles * n_stiles), block(256);
