@@ -264,10 +264,13 @@ def main() -> None:
             max_phi_err = float(e.item())
 
     if rank == 0:
+        headline = (args.config == "adult" and args.instances == 2560
+                    and args.background == 100)
         metric = (
             "explanations/sec (2560 inst, 100-sample background)"
-            if args.config == "adult"
-            else f"explanations/sec ({args.config} config)"
+            if headline
+            else f"explanations/sec ({args.config} config, "
+                 f"{args.instances} inst/GPU)"
         )
         result = {
             "metric": metric,
@@ -280,7 +283,7 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": (value / BASELINE_EXPL_PER_S
-                            if args.config == "adult" else None),
+                            if headline else None),
             "dtype": ("fp32" if args.dtype == "fp32" or device == "cpu"
                       else ("bf16x2 (hi+lo split, fp32-grade)"
                             if args.dtype == "bf16x2"
