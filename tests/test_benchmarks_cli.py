@@ -32,6 +32,10 @@ def test_bench_json_contract(tmp_path):
         assert key in obj, key
     assert obj["n_gpus"] == 1 and obj["scaling"] == "weak"
     assert obj["config"]["global_batch"] == 8
+    # a reduced instance count must NOT claim the headline metric/baseline
+    assert "2560 inst" not in obj["metric"]
+    assert obj["vs_baseline"] is None
+    assert "max_phi_err_vs_fp64" in obj
 
 
 def test_pool_cli_writes_results(tmp_path):
