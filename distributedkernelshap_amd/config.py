@@ -82,7 +82,10 @@ class KernelConfig:
     l1_device: bool = True
     # torch-module predictor path (mlp/resnet configs): bf16 autocast around
     # the module forward ("bf16") and channels-last weight layout for conv
-    # nets — the predict-bound configs' main levers on MI355X matrix cores
+    # nets — the predict-bound configs' main levers on MI355X matrix cores.
+    # channels_last only applies WITH bf16 autocast: MIOpen's fp32 NCHW
+    # path measured faster than NHWC (5.4 vs 3.2 expl/s on resnet), while
+    # bf16 NHWC wins (14.4 vs 12.4)
     module_autocast: str = "off"  # off | bf16
     module_channels_last: bool = True
 

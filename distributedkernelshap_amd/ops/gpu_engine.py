@@ -99,9 +99,12 @@ class GpuKernelShap:
             tm = getattr(engine.predictor, "torch_module", None)
             if tm is not None:
                 self.module = tm().to(self.device)
-                if engine.kernels.module_channels_last:
-                    # conv nets: NHWC weights select MIOpen's fast paths;
-                    # no-op for pure-linear modules
+                if (engine.kernels.module_channels_last
+                        and engine.kernels.module_autocast == "bf16"):
+                    # conv nets: NHWC weights select MIOpen's fast bf16
+                    # paths (measured: resnet bf16 12.4 -> 14.4 expl/s, but
+                    # fp32 NCHW is FASTER than NHWC — 5.4 vs 3.2 — so the
+                    # layout switch is tied to autocast)
                     try:
                         self.module = self.module.to(
                             memory_format=t.channels_last
