@@ -506,11 +506,16 @@ class GpuKernelShap:
         colg = t.tensor(
             vmap[self.engine._col_group].astype(np.int32), device=self.device
         )
+        # under bf16 autocast the synth kernel emits bf16 directly: halves
+        # the perturbation-tensor write traffic and removes the whole-tensor
+        # fp32->bf16 cast pass (mlp profile: 12.7 + ~6 ms per step)
+        sdt = (t.bfloat16 if self.engine.kernels.module_autocast == "bf16"
+               else t.float32)
         rows_per_inst = s * self.N
         if rows_per_inst <= chunk_rows:
             # pack g instances per call
             g_inst = max(1, min(b, chunk_rows // rows_per_inst))
-            buf = self._buf("synth", (g_inst * rows_per_inst, self.D))
+            buf = self._buf("synth", (g_inst * rows_per_inst, self.D), sdt)
             for lo in range(0, b, g_inst):
                 hi = min(lo + g_inst, b)
                 rows = (hi - lo) * rows_per_inst
@@ -524,7 +529,7 @@ class GpuKernelShap:
                 )
         else:
             s_chunk = max(1, chunk_rows // self.N)
-            buf = self._buf("synth", (s_chunk * self.N, self.D))
+            buf = self._buf("synth", (s_chunk * self.N, self.D), sdt)
             for bi in range(b):
                 for lo in range(0, s, s_chunk):
                     hi = min(lo + s_chunk, s)

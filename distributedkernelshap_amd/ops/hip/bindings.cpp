@@ -17,8 +17,8 @@ extern "C" int launch_fused_predict_linear(
 
 extern "C" void launch_synth_chunk(
     const uint8_t* masks, const float* x, const float* bg, const int* col_group,
-    float* out, int S, int M, int N, int D, int b_lo, int b_hi, int s_lo,
-    int s_hi, hipStream_t stream);
+    void* out, int out_bf16, int S, int M, int N, int D, int b_lo, int b_hi,
+    int s_lo, int s_hi, hipStream_t stream);
 
 // bf16 device pointers are opaque 16-bit words host-side (extern "C")
 extern "C" void launch_pack_masks(
@@ -123,9 +123,15 @@ void synth_chunk(
     int N = bg.size(0), D = bg.size(1);
     TORCH_CHECK(out.size(0) == (b_hi - b_lo) * (s_hi - s_lo) * N && out.size(1) == D,
                 "out shape");
+    const bool bf16 = out.dtype() == torch::kBFloat16;
+    TORCH_CHECK(bf16 || out.dtype() == torch::kFloat32,
+                "out must be float32 or bfloat16");
     launch_synth_chunk(
         masks.data_ptr<uint8_t>(), x.data_ptr<float>(), bg.data_ptr<float>(),
-        col_group.data_ptr<int>(), out.data_ptr<float>(), S, M, N, D,
+        col_group.data_ptr<int>(),
+        bf16 ? (void*)out.data_ptr<at::BFloat16>()
+             : (void*)out.data_ptr<float>(),
+        bf16 ? 1 : 0, S, M, N, D,
         (int)b_lo, (int)b_hi, (int)s_lo, (int)s_hi, current_stream());
 }
 

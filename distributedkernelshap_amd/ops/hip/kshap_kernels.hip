@@ -849,12 +849,13 @@ extern "C" void launch_build_diff_bf16(
 // Coalesced over d; one workgroup covers one (s, n) row-pair block.
 // ------------------------------------------------------------------------- //
 
+template <typename T>
 __global__ void synth_chunk_kernel(
     const uint8_t* __restrict__ masks,   // (B, S, M)
     const float* __restrict__ x,         // (B, D)
     const float* __restrict__ bg,        // (N, D)
     const int* __restrict__ col_group,   // (D)
-    float* __restrict__ out,             // ((b_hi-b_lo)*(s_hi-s_lo)*N, D)
+    T* __restrict__ out,                 // ((b_hi-b_lo)*(s_hi-s_lo)*N, D)
     int S, int M, int N, int D, int b_lo, int b_hi, int s_lo, int s_hi)
 {
     const int srange = s_hi - s_lo;
@@ -867,21 +868,33 @@ __global__ void synth_chunk_kernel(
     const uint8_t* mrow = masks + ((size_t)b * S + s) * M;
     const float* xrow = x + (size_t)b * D;
     const float* brow = bg + (size_t)n * D;
-    float* orow = out + (size_t)row * D;
+    T* orow = out + (size_t)row * D;
     for (int d = threadIdx.x; d < D; d += blockDim.x) {
-        orow[d] = mrow[col_group[d]] ? xrow[d] : brow[d];
+        orow[d] = (T)(mrow[col_group[d]] ? xrow[d] : brow[d]);
     }
 }
 
 extern "C" void launch_synth_chunk(
     const uint8_t* masks, const float* x, const float* bg, const int* col_group,
-    float* out, int S, int M, int N, int D, int b_lo, int b_hi, int s_lo,
-    int s_hi, hipStream_t stream)
+    void* out, int out_bf16, int S, int M, int N, int D, int b_lo, int b_hi,
+    int s_lo, int s_hi, hipStream_t stream)
 {
     size_t nrows = (size_t)(b_hi - b_lo) * (s_hi - s_lo) * N;
     int threads = D >= 256 ? 256 : (D >= 64 ? 64 : 32);
-    synth_chunk_kernel<<<dim3((unsigned)nrows), dim3(threads), 0, stream>>>(
-        masks, x, bg, col_group, out, S, M, N, D, b_lo, b_hi, s_lo, s_hi);
+    // bf16 output feeds autocast modules directly: halves the synth write
+    // traffic AND removes the separate fp32->bf16 cast pass torch would run
+    // over the whole perturbation tensor (mlp profile: 12.7 ms synth +
+    // 18 ms elementwise per step)
+    if (out_bf16)
+        synth_chunk_kernel<__bf16><<<dim3((unsigned)nrows), dim3(threads), 0,
+                                     stream>>>(
+            masks, x, bg, col_group, (__bf16*)out, S, M, N, D, b_lo, b_hi,
+            s_lo, s_hi);
+    else
+        synth_chunk_kernel<float><<<dim3((unsigned)nrows), dim3(threads), 0,
+                                    stream>>>(
+            masks, x, bg, col_group, (float*)out, S, M, N, D, b_lo, b_hi,
+            s_lo, s_hi);
 }
 
 // ------------------------------------------------------------------------- //
